@@ -1,0 +1,129 @@
+"""Driver benchmark contract (see BASELINE.json / BASELINE.md).
+
+Headline config: ResNet-18 / CIFAR-10-shaped synthetic data, repetition code r=3 with
+s=1 rev_grad adversary, colocated topology on N GPUs (weak scaling: each GPU computes
+r=3 group batches of 128 images per step; N groups total, so the DISTINCT trained
+images per step = N*128 — redundant compute is the cost of the code and is not
+counted in the throughput metric).
+
+Launched by the driver as
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+(single process, no torchrun, for N=1).  Rank 0 prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=128)
+    p.add_argument("--network", type=str, default="ResNet18")
+    p.add_argument("--dataset", type=str, default="Cifar10")
+    p.add_argument("--approach", type=str, default="maj_vote")
+    p.add_argument("--mode", type=str, default="maj_vote")
+    p.add_argument("--group-size", type=int, default=3)
+    p.add_argument("--worker-fail", type=int, default=1)
+    p.add_argument("--err-mode", type=str, default="rev_grad")
+    p.add_argument("--dtype", type=str, default="bf16")
+    p.add_argument("--device", type=str, default="auto")
+    args = p.parse_args()
+
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+
+    cfg = Config(
+        network=args.network,
+        dataset=args.dataset,
+        batch_size=args.batch_size,
+        approach=args.approach,
+        mode=args.mode,
+        group_size=args.group_size,
+        worker_fail=args.worker_fail,
+        err_mode=args.err_mode,
+        dtype=args.dtype,
+        device=args.device,
+        max_steps=args.steps + args.warmup + 10,
+        eval_freq=0,
+        log_dir="",
+        train_dir="gpurun_out/bench_ckpt",
+        topology="colocated",
+        deterministic=True,
+    )
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    on_gpu = t.device.type == "cuda"
+
+    for _ in range(args.warmup):
+        t.train_step()
+
+    t.comm.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        t.train_step()
+    t.comm.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (slowest rank defines the job step time)
+    el = torch.tensor([elapsed], dtype=torch.float64)
+    if t.comm.distributed:
+        el_dev = el.to(t.device) if t.comm.backend == "nccl" else el
+        t.comm.all_reduce(el_dev, op="max")
+        elapsed = float(el_dev[0])
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    # distinct images per step: one group-batch per group, G = world groups
+    distinct_per_step = world * args.batch_size
+    images_per_sec = distinct_per_step * args.steps / elapsed
+
+    if rank == 0:
+        result = {
+            "metric": "images/sec (effective), ResNet-18 CIFAR-10, repetition r=3, s=1 adversary",
+            "value": round(images_per_sec, 2),
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.network,
+                "global_batch": distinct_per_step,
+                "images_processed_per_step": world * args.group_size * args.batch_size,
+                "seq_len": None,
+                "input": "3x32x32" if args.dataset == "Cifar10" else args.dataset,
+                "parallelism": f"coded-dp{world}(repetition r={args.group_size}, s={args.worker_fail}, {args.err_mode})",
+                "topology": "colocated",
+                "optimizer": "fused SGD momentum=0.5",
+            },
+        }
+        print(json.dumps(result), flush=True)
+    t.close()
+    if t.comm.distributed:
+        t.comm.shutdown()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,78 @@
+"""Typed run configuration + CLI with the reference's user-facing flag names
+(/root/reference/src/distributed_nn.py:23-77, README.md:104-121) plus MI355X-native
+extensions (topology, dtype, workers-per-rank, vote-atol).
+"""
+from __future__ import annotations
+
+import argparse
+from dataclasses import dataclass, field, fields
+
+
+@dataclass
+class Config:
+    # reference-parity flags
+    batch_size: int = 128
+    test_batch_size: int = 100
+    max_steps: int = 10000
+    epochs: int = 100
+    lr: float = 0.01
+    momentum: float = 0.5
+    seed: int = 1
+    network: str = "LeNet"            # LeNet|FC|ResNet18|ResNet34|ResNet50|ResNet101|ResNet152|VGG11|VGG13|VGG16|VGG19
+    mode: str = "normal"              # normal|geometric_median|krum|maj_vote|cyclic (aggregation rule)
+    dataset: str = "MNIST"            # MNIST|Cifar10|ImageNetSynthetic
+    comm_type: str = "Bcast"          # accepted for parity; collectives are always fused (README.md:111)
+    err_mode: str = "rev_grad"        # rev_grad|constant|random|gauss|none
+    approach: str = "maj_vote"        # baseline|maj_vote|cyclic
+    num_aggregate: int = 5            # parity flag (unused, as in reference)
+    eval_freq: int = 50
+    train_dir: str = "output/models/"
+    adversarial: int = 1              # parity flag (err magnitude switch; reference hardcodes -100)
+    worker_fail: int = 2
+    group_size: int = 5
+    compress_grad: str = "none"       # none|bf16 (blosc replaced by GPU dtype cast)
+    checkpoint_step: int = 0
+
+    # MI355X-native extensions
+    topology: str = "colocated"       # colocated (all ranks compute) | ps (rank0 = parameter server)
+    workers_per_rank: int = 0         # 0 = derive from approach (r for maj_vote, 1 otherwise)
+    vote_atol: float = 0.0            # 0.0 = bitwise-equality vote (reference semantics)
+    dtype: str = "bf16"               # bf16|fp32 compute dtype (grads/aggregation always fp32)
+    device: str = "auto"              # auto|cuda|cpu
+    deterministic: bool = True
+    log_dir: str = "output/logs/"
+
+    def sanity(self):
+        if self.approach == "maj_vote" and self.group_size < 1:
+            raise ValueError("group-size must be >= 1 for maj_vote")
+        if self.approach == "cyclic" and self.worker_fail < 1:
+            raise ValueError("cyclic approach needs worker-fail >= 1")
+
+
+def add_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
+    for f in fields(Config):
+        flag = "--" + f.name.replace("_", "-")
+        if f.type == "bool" or isinstance(f.default, bool):
+            parser.add_argument(flag, type=lambda s: s.lower() in ("1", "true", "yes"),
+                                default=f.default)
+        else:
+            parser.add_argument(flag, type=type(f.default), default=f.default)
+    # accepted-for-parity flags with no framework meaning
+    parser.add_argument("--hostfile", type=str, default="", help="accepted for CLI parity; ranks come from torchrun env")
+    parser.add_argument("--no-cuda", action="store_true", default=False)
+    parser.add_argument("--log-interval", type=int, default=10)
+    return parser
+
+
+def from_args(args: argparse.Namespace) -> Config:
+    cfg = Config(**{f.name: getattr(args, f.name) for f in fields(Config)})
+    if getattr(args, "no_cuda", False):
+        cfg.device = "cpu"
+    cfg.sanity()
+    return cfg
+
+
+def parse_cli(argv=None) -> Config:
+    p = argparse.ArgumentParser(description="draco_amd distributed training")
+    add_args(p)
+    return from_args(p.parse_args(argv))

@@ -1,0 +1,72 @@
+"""Deterministic synthetic datasets (no network access in this environment).
+
+Index-addressed like the reference's deterministic global batches
+(/root/reference/src/datasets/utils.py:21-29 `get_batch`): batch contents are a pure
+function of (seed, indices), so
+  * repetition-code group members on DIFFERENT ranks draw bit-identical batches
+    (the group-seed mechanism of util.py:69-97 / rep_worker.py:89), and
+  * cyclic-code sub-batches are globally addressable slices, exactly like
+    cyclic_worker.py:94.
+
+The task is learnable: labels are drawn from the index stream and images are
+class-mean + noise (fixed class means), so convergence-under-attack experiments mirror
+the reference methodology without torchvision downloads.
+"""
+from __future__ import annotations
+
+import torch
+
+from ..models import dataset_shape
+
+
+class SyntheticClassification:
+    """Infinite deterministic class-conditional Gaussian image stream."""
+
+    def __init__(self, dataset: str, device: torch.device, seed: int = 1234, noise: float = 1.0, dtype=torch.float32):
+        c, h, w, classes = dataset_shape(dataset)
+        self.shape = (c, h, w)
+        self.classes = classes
+        self.device = device
+        self.noise = noise
+        self.dtype = dtype
+        g = torch.Generator(device="cpu")
+        g.manual_seed(seed)
+        # fixed class means, modest separation so training has to work for it
+        self.means = (torch.randn(classes, c, h, w, generator=g) * 0.7).to(device=device, dtype=dtype)
+
+    def get_batch(self, start: int, batch: int):
+        """Deterministic batch for global sample indices [start, start+batch)."""
+        g = torch.Generator(device="cpu")
+        g.manual_seed(0x9E3779B9 ^ (start & 0xFFFFFFFF) ^ ((start >> 32) << 1))
+        y = torch.randint(0, self.classes, (batch,), generator=g)
+        x = torch.randn(batch, *self.shape, generator=g) * self.noise
+        x = x.to(device=self.device, dtype=self.dtype) + self.means[y.to(self.device)]
+        return x, y.to(self.device)
+
+
+class GroupBatchSource:
+    """Batch stream keyed by (group, step): all members of a group draw the identical
+    batch (repetition code), with disjoint data across groups and steps."""
+
+    def __init__(self, data: SyntheticClassification, batch_size: int, n_groups: int):
+        self.data = data
+        self.B = batch_size
+        self.G = n_groups
+
+    def batch_for(self, group: int, step: int):
+        start = (step * self.G + group) * self.B
+        return self.data.get_batch(start, self.B)
+
+
+class GlobalBatchSource:
+    """Cyclic-code stream: a global batch of n*B samples per step, sub-batch j is the
+    j-th slice (cyclic_worker.py:94,122-126)."""
+
+    def __init__(self, data: SyntheticClassification, batch_size: int, n_workers: int):
+        self.data = data
+        self.B = batch_size
+        self.n = n_workers
+
+    def sub_batch(self, sub_index: int, step: int):
+        start = (step * self.n + sub_index) * self.B
+        return self.data.get_batch(start, self.B)

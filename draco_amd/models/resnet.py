@@ -1,0 +1,138 @@
+"""CIFAR ResNet-18/34/50/101/152 (parity with /root/reference/src/model_ops/resnet.py:14-113)
+plus an ImageNet-geometry ResNet-50 for the 224x224 synthetic config (BASELINE config 5).
+
+Written fresh in idiomatic PyTorch; runs on PyTorch-ROCm (MIOpen/hipBLASLt kernels).
+"""
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, in_planes, planes, stride=1):
+        super().__init__()
+        self.conv1 = nn.Conv2d(in_planes, planes, 3, stride=stride, padding=1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = nn.Conv2d(planes, planes, 3, stride=1, padding=1, bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.shortcut = nn.Sequential()
+        if stride != 1 or in_planes != planes * self.expansion:
+            self.shortcut = nn.Sequential(
+                nn.Conv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
+                nn.BatchNorm2d(planes * self.expansion),
+            )
+
+    def forward(self, x):
+        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        out = out + self.shortcut(x)
+        return F.relu(out)
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, in_planes, planes, stride=1):
+        super().__init__()
+        self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.conv3 = nn.Conv2d(planes, planes * self.expansion, 1, bias=False)
+        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.shortcut = nn.Sequential()
+        if stride != 1 or in_planes != planes * self.expansion:
+            self.shortcut = nn.Sequential(
+                nn.Conv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
+                nn.BatchNorm2d(planes * self.expansion),
+            )
+
+    def forward(self, x):
+        out = F.relu(self.bn1(self.conv1(x)))
+        out = F.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        out = out + self.shortcut(x)
+        return F.relu(out)
+
+
+class ResNet(nn.Module):
+    """CIFAR geometry: 3x3 stem, 4 stages at 32x32 input."""
+
+    def __init__(self, block, num_blocks, num_classes=10, in_channels=3):
+        super().__init__()
+        self.in_planes = 64
+        self.conv1 = nn.Conv2d(in_channels, 64, 3, stride=1, padding=1, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.layer1 = self._make_layer(block, 64, num_blocks[0], 1)
+        self.layer2 = self._make_layer(block, 128, num_blocks[1], 2)
+        self.layer3 = self._make_layer(block, 256, num_blocks[2], 2)
+        self.layer4 = self._make_layer(block, 512, num_blocks[3], 2)
+        self.linear = nn.Linear(512 * block.expansion, num_classes)
+
+    def _make_layer(self, block, planes, n, stride):
+        layers = []
+        for s in [stride] + [1] * (n - 1):
+            layers.append(block(self.in_planes, planes, s))
+            self.in_planes = planes * block.expansion
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.layer1(out)
+        out = self.layer2(out)
+        out = self.layer3(out)
+        out = self.layer4(out)
+        out = F.adaptive_avg_pool2d(out, 1).flatten(1)
+        return self.linear(out)
+
+
+class ResNetImageNet(nn.Module):
+    """ImageNet geometry (7x7 stem + maxpool) for 224x224 synthetic runs."""
+
+    def __init__(self, block, num_blocks, num_classes=1000, in_channels=3):
+        super().__init__()
+        self.in_planes = 64
+        self.conv1 = nn.Conv2d(in_channels, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.layer1 = self._make_layer(block, 64, num_blocks[0], 1)
+        self.layer2 = self._make_layer(block, 128, num_blocks[1], 2)
+        self.layer3 = self._make_layer(block, 256, num_blocks[2], 2)
+        self.layer4 = self._make_layer(block, 512, num_blocks[3], 2)
+        self.linear = nn.Linear(512 * block.expansion, num_classes)
+
+    _make_layer = ResNet._make_layer
+
+    def forward(self, x):
+        out = self.maxpool(F.relu(self.bn1(self.conv1(x))))
+        out = self.layer1(out)
+        out = self.layer2(out)
+        out = self.layer3(out)
+        out = self.layer4(out)
+        out = F.adaptive_avg_pool2d(out, 1).flatten(1)
+        return self.linear(out)
+
+
+def ResNet18(num_classes=10, in_channels=3):
+    return ResNet(BasicBlock, [2, 2, 2, 2], num_classes, in_channels)
+
+
+def ResNet34(num_classes=10, in_channels=3):
+    return ResNet(BasicBlock, [3, 4, 6, 3], num_classes, in_channels)
+
+
+def ResNet50(num_classes=10, in_channels=3):
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes, in_channels)
+
+
+def ResNet101(num_classes=10, in_channels=3):
+    return ResNet(Bottleneck, [3, 4, 23, 3], num_classes, in_channels)
+
+
+def ResNet152(num_classes=10, in_channels=3):
+    return ResNet(Bottleneck, [3, 8, 36, 3], num_classes, in_channels)
+
+
+def ResNet50ImageNet(num_classes=1000, in_channels=3):
+    return ResNetImageNet(Bottleneck, [3, 4, 6, 3], num_classes, in_channels)

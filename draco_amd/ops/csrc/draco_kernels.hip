@@ -1,0 +1,519 @@
+// draco_amd HIP/CDNA4 kernels — MI355X (gfx950) native.
+//
+// Every coding / aggregation / optimizer hot spot of the framework runs through the
+// kernels in this file (SURVEY.md §2.5 K1-K12 inventory; reference semantics cited per
+// kernel).  All of them are HBM-bandwidth-bound elementwise/reduction kernels over the
+// flat gradient space (d up to ~10^7 fp32 per model replica), so the design rules are
+// the memory ones from the CDNA4 guide: 256-thread blocks (4 waves of 64), float4
+// (16 B/lane) vectorized access, grid-stride loops capped at ~2048 blocks, wave-level
+// __shfl reductions (wave = 64 lanes on CDNA4, not 32).
+//
+// Build: hipcc --offload-arch=gfx950 (tools/build_ext.py); loaded as torch extension
+// draco_amd._hip_ops.  No CUDA path, no hipify — HIP-native source.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#define WAVE 64
+#define NTHREADS 256
+
+static inline int n_blocks(long work, int per_block) {
+  long b = (work + per_block - 1) / per_block;
+  if (b > 2048) b = 2048;
+  if (b < 1) b = 1;
+  return (int)b;
+}
+
+#define CHECK_IN(t) \
+  TORCH_CHECK(t.is_cuda(), #t " must be on GPU"); \
+  TORCH_CHECK(t.is_contiguous(), #t " must be contiguous")
+
+static inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+// --------------------------------------------------------------------------- SGD
+// Fused flat SGD + momentum/nesterov/weight-decay (reference semantics:
+// optim/sgd_modified.py:53-89 including the first-step dampening quirk).
+// One launch for the whole parameter space vs the reference's per-tensor loop.
+template <bool MOM, bool NESTEROV>
+__global__ void k_sgd(float4 *__restrict__ p, const float4 *__restrict__ g,
+                      float4 *__restrict__ buf, long n4, float lr, float momentum,
+                      float grad_scale /* 1-dampening, or 1 on first step */, float wd) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  for (; i < n4; i += stride) {
+    float4 pv = p[i];
+    float4 gv = g[i];
+    float dp[4] = {gv.x, gv.y, gv.z, gv.w};
+    float pp[4] = {pv.x, pv.y, pv.z, pv.w};
+    if (wd != 0.f) {
+#pragma unroll
+      for (int c = 0; c < 4; ++c) dp[c] = fmaf(wd, pp[c], dp[c]);
+    }
+    if (MOM) {
+      float4 bv = buf[i];
+      float bb[4] = {bv.x, bv.y, bv.z, bv.w};
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        bb[c] = fmaf(momentum, bb[c], grad_scale * dp[c]);
+        dp[c] = NESTEROV ? fmaf(momentum, bb[c], dp[c]) : bb[c];
+      }
+      buf[i] = make_float4(bb[0], bb[1], bb[2], bb[3]);
+#pragma unroll
+      for (int c = 0; c < 4; ++c) pp[c] = fmaf(-lr, dp[c], pp[c]);
+    } else {
+#pragma unroll
+      for (int c = 0; c < 4; ++c) pp[c] = fmaf(-lr, dp[c], pp[c]);
+    }
+    p[i] = make_float4(pp[0], pp[1], pp[2], pp[3]);
+  }
+}
+
+void fused_sgd_step(torch::Tensor param, torch::Tensor grad, torch::Tensor buf,
+                    double lr, double momentum, double dampening, double weight_decay,
+                    bool nesterov, bool first_step) {
+  CHECK_IN(param); CHECK_IN(grad);
+  TORCH_CHECK(param.numel() % 4 == 0, "flat param length must be a multiple of 4");
+  long n4 = param.numel() / 4;
+  int blocks = n_blocks(n4, NTHREADS);
+  bool mom = momentum != 0.0;
+  float gscale = first_step ? 1.f : (1.f - (float)dampening);
+  auto pp = (float4 *)param.data_ptr<float>();
+  auto gg = (const float4 *)grad.data_ptr<float>();
+  float4 *bb = nullptr;
+  if (mom) { CHECK_IN(buf); bb = (float4 *)buf.data_ptr<float>(); }
+  hipStream_t s = cur_stream();
+  if (mom && nesterov)
+    hipLaunchKernelGGL((k_sgd<true, true>), dim3(blocks), dim3(NTHREADS), 0, s,
+                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay);
+  else if (mom)
+    hipLaunchKernelGGL((k_sgd<true, false>), dim3(blocks), dim3(NTHREADS), 0, s,
+                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay);
+  else
+    hipLaunchKernelGGL((k_sgd<false, false>), dim3(blocks), dim3(NTHREADS), 0, s,
+                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay);
+}
+
+// --------------------------------------------------------------------------- Adam
+// Fused flat Adam/AMSGrad (reference: optim/adam_modified.py:32-93).
+template <bool AMS>
+__global__ void k_adam(float4 *__restrict__ p, const float4 *__restrict__ g,
+                       float4 *__restrict__ m, float4 *__restrict__ v,
+                       float4 *__restrict__ vmax, long n4, float lr_t, float beta1,
+                       float beta2, float eps, float wd) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  for (; i < n4; i += stride) {
+    float4 pv = p[i], gv = g[i], mv = m[i], vv = v[i];
+    float pp[4] = {pv.x, pv.y, pv.z, pv.w};
+    float dg[4] = {gv.x, gv.y, gv.z, gv.w};
+    float mm[4] = {mv.x, mv.y, mv.z, mv.w};
+    float ss[4] = {vv.x, vv.y, vv.z, vv.w};
+    float dn[4];
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      if (wd != 0.f) dg[c] = fmaf(wd, pp[c], dg[c]);
+      mm[c] = fmaf(beta1, mm[c], (1.f - beta1) * dg[c]);
+      ss[c] = fmaf(beta2, ss[c], (1.f - beta2) * dg[c] * dg[c]);
+      dn[c] = ss[c];
+    }
+    if (AMS) {
+      float4 xv = vmax[i];
+      float xx[4] = {xv.x, xv.y, xv.z, xv.w};
+#pragma unroll
+      for (int c = 0; c < 4; ++c) { xx[c] = fmaxf(xx[c], ss[c]); dn[c] = xx[c]; }
+      vmax[i] = make_float4(xx[0], xx[1], xx[2], xx[3]);
+    }
+#pragma unroll
+    for (int c = 0; c < 4; ++c)
+      pp[c] -= lr_t * mm[c] / (sqrtf(dn[c]) + eps);
+    m[i] = make_float4(mm[0], mm[1], mm[2], mm[3]);
+    v[i] = make_float4(ss[0], ss[1], ss[2], ss[3]);
+    p[i] = make_float4(pp[0], pp[1], pp[2], pp[3]);
+  }
+}
+
+void fused_adam_step(torch::Tensor param, torch::Tensor grad, torch::Tensor exp_avg,
+                     torch::Tensor exp_avg_sq, torch::Tensor max_exp_avg_sq, long step,
+                     double lr, double beta1, double beta2, double eps,
+                     double weight_decay, bool amsgrad) {
+  CHECK_IN(param); CHECK_IN(grad); CHECK_IN(exp_avg); CHECK_IN(exp_avg_sq);
+  long n4 = param.numel() / 4;
+  int blocks = n_blocks(n4, NTHREADS);
+  double bc1 = 1.0 - pow(beta1, (double)step);
+  double bc2 = 1.0 - pow(beta2, (double)step);
+  float lr_t = (float)(lr * sqrt(bc2) / bc1);
+  hipStream_t s = cur_stream();
+  auto pp = (float4 *)param.data_ptr<float>();
+  auto gg = (const float4 *)grad.data_ptr<float>();
+  auto mm = (float4 *)exp_avg.data_ptr<float>();
+  auto vv = (float4 *)exp_avg_sq.data_ptr<float>();
+  if (amsgrad) {
+    CHECK_IN(max_exp_avg_sq);
+    hipLaunchKernelGGL((k_adam<true>), dim3(blocks), dim3(NTHREADS), 0, s, pp, gg, mm, vv,
+                       (float4 *)max_exp_avg_sq.data_ptr<float>(), n4, lr_t,
+                       (float)beta1, (float)beta2, (float)eps, (float)weight_decay);
+  } else {
+    hipLaunchKernelGGL((k_adam<false>), dim3(blocks), dim3(NTHREADS), 0, s, pp, gg, mm, vv,
+                       nullptr, n4, lr_t, (float)beta1, (float)beta2, (float)eps,
+                       (float)weight_decay);
+  }
+}
+
+// --------------------------------------------------------------------------- inject
+// Adversary injection y = a*y + b covers every reference err_simulation mode
+// (model_ops/utils.py:6-23): rev_grad (a=-100,b=0), rev_grad cyclic (a=-99,b=0),
+// constant (a=0,b=-100), constant cyclic (a=1,b=-100).
+__global__ void k_axpb(float4 *__restrict__ y, long n4, float a, float b) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  for (; i < n4; i += stride) {
+    float4 v = y[i];
+    y[i] = make_float4(fmaf(a, v.x, b), fmaf(a, v.y, b), fmaf(a, v.z, b), fmaf(a, v.w, b));
+  }
+}
+
+void inject(torch::Tensor grad, std::string mode, bool cyclic) {
+  CHECK_IN(grad);
+  const float ADV = -100.f;
+  float a, b;
+  if (mode == "rev_grad") { a = cyclic ? (1.f + ADV) : ADV; b = 0.f; }
+  else if (mode == "constant") { a = cyclic ? 1.f : 0.f; b = ADV; }
+  else TORCH_CHECK(false, "inject: unknown mode ", mode);
+  long n4 = grad.numel() / 4;
+  hipLaunchKernelGGL(k_axpb, dim3(n_blocks(n4, NTHREADS)), dim3(NTHREADS), 0, cur_stream(),
+                     (float4 *)grad.data_ptr<float>(), n4, a, b);
+}
+
+// --------------------------------------------------------------------------- vote
+// Majority-vote equality test (reference: np.array_equal per member,
+// rep_master.py:154-168).  out[pair] preset to 1; any mismatching element anywhere in
+// the shard stores 0 (benign race — no reduction needed, one pass at full bandwidth).
+__global__ void k_rows_equal(const float4 *__restrict__ x, const long *__restrict__ ai,
+                             const long *__restrict__ bi, unsigned char *__restrict__ out,
+                             long d4, long stride4, float atol) {
+  int pair = blockIdx.y;
+  const float4 *ra = x + ai[pair] * stride4;
+  const float4 *rb = x + bi[pair] * stride4;
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  bool neq = false;
+  for (; i < d4; i += stride) {
+    float4 a = ra[i], b = rb[i];
+    neq |= !(fabsf(a.x - b.x) <= atol) | !(fabsf(a.y - b.y) <= atol) |
+           !(fabsf(a.z - b.z) <= atol) | !(fabsf(a.w - b.w) <= atol);
+  }
+  // one store per wave suffices when any lane saw a mismatch (benign race on out)
+  if (__any(neq) && (threadIdx.x & (WAVE - 1)) == 0) out[pair] = 0;
+}
+
+torch::Tensor rows_equal(torch::Tensor x, torch::Tensor a_idx, torch::Tensor b_idx,
+                         double atol) {
+  CHECK_IN(x); CHECK_IN(a_idx); CHECK_IN(b_idx);
+  TORCH_CHECK(x.dim() == 2, "rows_equal: x must be 2D");
+  long k = a_idx.numel();
+  long d = x.size(1);
+  TORCH_CHECK(d % 4 == 0, "rows_equal: row length must be a multiple of 4");
+  auto out = torch::ones({k}, torch::dtype(torch::kUInt8).device(x.device()));
+  if (d == 0 || k == 0) return out;
+  dim3 grid(n_blocks(d / 4, NTHREADS), (unsigned)k);
+  hipLaunchKernelGGL(k_rows_equal, grid, dim3(NTHREADS), 0, cur_stream(),
+                     (const float4 *)x.data_ptr<float>(), a_idx.data_ptr<long>(),
+                     b_idx.data_ptr<long>(), out.data_ptr<unsigned char>(), d / 4, d / 4,
+                     (float)atol);
+  return out;
+}
+
+// --------------------------------------------------------------------------- combine
+// Generic weighted row combination out[j] = sum_i w[i] * x[rows[i]][j].
+// Serves: mean of group winners (K8), sum_rows, cyclic encode per-plane and the final
+// decode recombination Re(v @ R) (K1/K5) — the n<=2*64 coefficients live in L2 and are
+// re-read per thread (negligible vs the d-wide streams).
+__global__ void k_combine(const float4 *__restrict__ x, const long *__restrict__ rows,
+                          const float *__restrict__ w, int m, float4 *__restrict__ out,
+                          long d4, long stride4) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  for (; i < d4; i += stride) {
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+    for (int r = 0; r < m; ++r) {
+      float4 v = x[rows[r] * stride4 + i];
+      float wr = w[r];
+      acc[0] = fmaf(wr, v.x, acc[0]);
+      acc[1] = fmaf(wr, v.y, acc[1]);
+      acc[2] = fmaf(wr, v.z, acc[2]);
+      acc[3] = fmaf(wr, v.w, acc[3]);
+    }
+    out[i] = make_float4(acc[0], acc[1], acc[2], acc[3]);
+  }
+}
+
+static void launch_combine(torch::Tensor x, torch::Tensor rows, torch::Tensor w,
+                           torch::Tensor out, long stride_elems) {
+  long d = out.numel();
+  TORCH_CHECK(d % 4 == 0 && stride_elems % 4 == 0, "combine: 16-byte alignment required");
+  hipLaunchKernelGGL(k_combine, dim3(n_blocks(d / 4, NTHREADS)), dim3(NTHREADS), 0,
+                     cur_stream(), (const float4 *)x.data_ptr<float>(),
+                     rows.data_ptr<long>(), w.data_ptr<float>(), (int)rows.numel(),
+                     (float4 *)out.data_ptr<float>(), d / 4, stride_elems / 4);
+}
+
+void mean_rows(torch::Tensor x, torch::Tensor idx, torch::Tensor out) {
+  CHECK_IN(x); CHECK_IN(idx); CHECK_IN(out);
+  auto w = torch::full({idx.numel()}, 1.0 / (double)idx.numel(),
+                       torch::dtype(torch::kFloat32).device(x.device()));
+  launch_combine(x, idx, w, out, x.size(1));
+}
+
+void sum_rows(torch::Tensor x, torch::Tensor out) {
+  CHECK_IN(x); CHECK_IN(out);
+  auto idx = torch::arange(x.size(0), torch::dtype(torch::kInt64).device(x.device()));
+  auto w = torch::ones({x.size(0)}, torch::dtype(torch::kFloat32).device(x.device()));
+  launch_combine(x, idx, w, out, x.size(1));
+}
+
+// Cyclic encode (K1, cyclic_worker.py:172-176): one read of the (2s+1, d) sub-batch
+// gradients produces BOTH complex planes (re/im written in the same pass).
+__global__ void k_encode2(const float4 *__restrict__ g, const float *__restrict__ wre,
+                          const float *__restrict__ wim, int m, float4 *__restrict__ outre,
+                          float4 *__restrict__ outim, long d4, long stride4) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  for (; i < d4; i += stride) {
+    float re[4] = {0, 0, 0, 0}, im[4] = {0, 0, 0, 0};
+    for (int r = 0; r < m; ++r) {
+      float4 v = g[r * stride4 + i];
+      float a = wre[r], b = wim[r];
+      re[0] = fmaf(a, v.x, re[0]); im[0] = fmaf(b, v.x, im[0]);
+      re[1] = fmaf(a, v.y, re[1]); im[1] = fmaf(b, v.y, im[1]);
+      re[2] = fmaf(a, v.z, re[2]); im[2] = fmaf(b, v.z, im[2]);
+      re[3] = fmaf(a, v.w, re[3]); im[3] = fmaf(b, v.w, im[3]);
+    }
+    outre[i] = make_float4(re[0], re[1], re[2], re[3]);
+    outim[i] = make_float4(im[0], im[1], im[2], im[3]);
+  }
+}
+
+void cyclic_encode(torch::Tensor grads, torch::Tensor w_re, torch::Tensor w_im,
+                   torch::Tensor out) {
+  CHECK_IN(grads); CHECK_IN(out);
+  TORCH_CHECK(out.size(0) == 2, "cyclic_encode: out must be (2, d)");
+  long d = out.size(1);
+  long stride = grads.size(1);
+  hipLaunchKernelGGL(k_encode2, dim3(n_blocks(d / 4, NTHREADS)), dim3(NTHREADS), 0,
+                     cur_stream(), (const float4 *)grads.data_ptr<float>(),
+                     w_re.data_ptr<float>(), w_im.data_ptr<float>(), (int)grads.size(0),
+                     (float4 *)out.data_ptr<float>(),
+                     (float4 *)out.data_ptr<float>() + d / 4, d / 4, stride / 4);
+}
+
+void cyclic_recombine(torch::Tensor r_planes, torch::Tensor v_re, torch::Tensor v_im,
+                      torch::Tensor out) {
+  CHECK_IN(r_planes); CHECK_IN(out);
+  // rows (2i, 2i+1) are worker i's (re, im) planes; Re(v @ R) = sum vre*re - vim*im
+  long n = r_planes.size(0);
+  auto dev = r_planes.device();
+  auto rows = torch::arange(2 * n, torch::dtype(torch::kInt64).device(dev));
+  auto w = torch::stack({v_re, -v_im}, 1).reshape({2 * n}).contiguous().to(dev);
+  launch_combine(r_planes.view({2 * n, -1}), rows, w, out, r_planes.size(2));
+}
+
+// --------------------------------------------------------------------------- project
+// K3 (cyclic_master.py:154): partial projection proj[i] = sum_d R[i,d] * z[d] over the
+// local shard.  One block row per (worker, plane); wave shuffle reduce (64-lane) then
+// LDS across the block's 4 waves, one atomicAdd per block.
+__global__ void k_project(const float4 *__restrict__ r, const float4 *__restrict__ z,
+                          float *__restrict__ out, long d4, long stride4) {
+  int row = blockIdx.y;
+  const float4 *rr = r + row * stride4;
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  float acc = 0.f;
+  for (; i < d4; i += stride) {
+    float4 a = rr[i], b = z[i];
+    acc = fmaf(a.x, b.x, acc);
+    acc = fmaf(a.y, b.y, acc);
+    acc = fmaf(a.z, b.z, acc);
+    acc = fmaf(a.w, b.w, acc);
+  }
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) acc += __shfl_down(acc, off, WAVE);
+  __shared__ float warp_acc[NTHREADS / WAVE];
+  int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) warp_acc[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float tot = 0.f;
+    for (int wv = 0; wv < NTHREADS / WAVE; ++wv) tot += warp_acc[wv];
+    atomicAdd(&out[row], tot);
+  }
+}
+
+torch::Tensor cyclic_project(torch::Tensor r_planes, torch::Tensor z) {
+  CHECK_IN(r_planes); CHECK_IN(z);
+  long n = r_planes.size(0);
+  long d = r_planes.size(2);
+  auto out = torch::zeros({n, 2}, torch::dtype(torch::kFloat32).device(r_planes.device()));
+  if (d == 0) return out;
+  dim3 grid(n_blocks(d / 4 / 8, NTHREADS), (unsigned)(2 * n));
+  hipLaunchKernelGGL(k_project, grid, dim3(NTHREADS), 0, cur_stream(),
+                     (const float4 *)r_planes.data_ptr<float>(),
+                     (const float4 *)z.data_ptr<float>(), out.data_ptr<float>(), d / 4,
+                     d / 4);
+  return out;
+}
+
+// --------------------------------------------------------------------------- geomed
+// K6 (baseline_master.py:271-276 / hdmedians): per-(worker, layer-segment) partial
+// squared distances for the sharded Weiszfeld iteration.  Segment bounds live in LDS;
+// each thread keeps a running (segment, acc) and flushes on change via atomicAdd.
+__global__ void k_seg_sqdist(const float *__restrict__ x, const float *__restrict__ z,
+                             const long *__restrict__ seg, int L,
+                             float *__restrict__ out /* (P, L) */, long d, long stride) {
+  extern __shared__ long s_seg[];
+  for (int i = threadIdx.x; i <= L; i += blockDim.x) s_seg[i] = seg[i];
+  __syncthreads();
+  int p = blockIdx.y;
+  const float *row = x + p * stride;
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long gstride = gridDim.x * (long)blockDim.x;
+  int cur = -1;
+  float acc = 0.f;
+  for (; i < d; i += gstride) {
+    // binary search: segment l with seg[l] <= i < seg[l+1]
+    int lo = 0, hi = L - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (s_seg[mid] <= i) lo = mid; else hi = mid - 1;
+    }
+    float dlt = row[i] - z[i];
+    if (lo != cur) {
+      if (cur >= 0 && acc != 0.f) atomicAdd(&out[p * L + cur], acc);
+      cur = lo; acc = 0.f;
+    }
+    acc = fmaf(dlt, dlt, acc);
+  }
+  if (cur >= 0 && acc != 0.f) atomicAdd(&out[p * L + cur], acc);
+}
+
+torch::Tensor segment_sqdist(torch::Tensor x, torch::Tensor z, torch::Tensor seg) {
+  CHECK_IN(x); CHECK_IN(z); CHECK_IN(seg);
+  long P = x.size(0), d = x.size(1);
+  int L = (int)seg.numel() - 1;
+  auto out = torch::zeros({P, L}, torch::dtype(torch::kFloat32).device(x.device()));
+  if (d == 0) return out;
+  dim3 grid(n_blocks(d / 4, NTHREADS), (unsigned)P);
+  hipLaunchKernelGGL(k_seg_sqdist, grid, dim3(NTHREADS), (L + 1) * sizeof(long),
+                     cur_stream(), x.data_ptr<float>(), z.data_ptr<float>(),
+                     seg.data_ptr<long>(), L, out.data_ptr<float>(), d, d);
+  return out;
+}
+
+// Weiszfeld update z[j] = sum_p w[p, seg(j)] * x[p, j] (weights pre-normalised).
+__global__ void k_seg_wmean(const float *__restrict__ x, const float *__restrict__ w,
+                            const long *__restrict__ seg, int L, int P,
+                            float *__restrict__ out, long d, long stride) {
+  extern __shared__ long s_seg[];
+  for (int i = threadIdx.x; i <= L; i += blockDim.x) s_seg[i] = seg[i];
+  __syncthreads();
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long gstride = gridDim.x * (long)blockDim.x;
+  for (; i < d; i += gstride) {
+    int lo = 0, hi = L - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (s_seg[mid] <= i) lo = mid; else hi = mid - 1;
+    }
+    float acc = 0.f;
+    for (int p = 0; p < P; ++p) acc = fmaf(w[p * L + lo], x[p * stride + i], acc);
+    out[i] = acc;
+  }
+}
+
+void segment_weighted_mean(torch::Tensor x, torch::Tensor w, torch::Tensor seg,
+                           torch::Tensor out) {
+  CHECK_IN(x); CHECK_IN(w); CHECK_IN(seg); CHECK_IN(out);
+  long P = x.size(0), d = x.size(1);
+  int L = (int)seg.numel() - 1;
+  if (d == 0) return;
+  hipLaunchKernelGGL(k_seg_wmean, dim3(n_blocks(d, NTHREADS)), dim3(NTHREADS),
+                     (L + 1) * sizeof(long), cur_stream(), x.data_ptr<float>(),
+                     w.data_ptr<float>(), seg.data_ptr<long>(), L, (int)P,
+                     out.data_ptr<float>(), d, d);
+}
+
+// --------------------------------------------------------------------------- krum
+// K7 (baseline_master.py:278-296): per-segment Gram matrices for the pairwise
+// distance matrix.  Host pre-splits segments into <=CHUNK-element descriptors; a block
+// stages its chunk (P x CHUNK) through LDS and accumulates all P*P pair dots.
+#define GRAM_CHUNK 512
+__global__ void k_seg_gram(const float *__restrict__ x, const long *__restrict__ desc,
+                           /* desc: (nchunks, 3) = (seg, start, len) */
+                           int P, float *__restrict__ out /* (L, P, P) */, long stride) {
+  __shared__ float tile[32][GRAM_CHUNK / 4][4];  // [P][chunk] as float4 groups
+  int c = blockIdx.x;
+  long segid = desc[c * 3 + 0];
+  long start = desc[c * 3 + 1];
+  long len = desc[c * 3 + 2];
+  // cooperative load: threads sweep (p, e)
+  for (long t = threadIdx.x; t < (long)P * len; t += blockDim.x) {
+    int p = (int)(t / len);
+    long e = t % len;
+    tile[p][e / 4][e % 4] = x[p * stride + start + e];
+  }
+  __syncthreads();
+  float *base = out + segid * P * P;
+  for (int pair = threadIdx.x; pair < P * P; pair += blockDim.x) {
+    int a = pair / P, b = pair % P;
+    if (b < a) continue;  // symmetric: fill upper, mirror below
+    float acc = 0.f;
+    for (long e = 0; e < len; ++e)
+      acc = fmaf(tile[a][e / 4][e % 4], tile[b][e / 4][e % 4], acc);
+    atomicAdd(&base[a * P + b], acc);
+    if (a != b) atomicAdd(&base[b * P + a], acc);
+  }
+}
+
+torch::Tensor segment_gram(torch::Tensor x, torch::Tensor seg) {
+  CHECK_IN(x); CHECK_IN(seg);
+  long P = x.size(0), d = x.size(1);
+  int L = (int)seg.numel() - 1;
+  TORCH_CHECK(P <= 32, "segment_gram kernel supports P <= 32");
+  auto out = torch::zeros({L, P, P}, torch::dtype(torch::kFloat32).device(x.device()));
+  // host-side chunk descriptors
+  auto seg_c = seg.to(torch::kCPU);
+  auto sp = seg_c.data_ptr<long>();
+  std::vector<long> desc;
+  for (int l = 0; l < L; ++l) {
+    for (long s0 = sp[l]; s0 < sp[l + 1]; s0 += GRAM_CHUNK) {
+      desc.push_back(l);
+      desc.push_back(s0);
+      desc.push_back(std::min((long)GRAM_CHUNK, sp[l + 1] - s0));
+    }
+  }
+  if (desc.empty()) return out;
+  auto desc_t = torch::from_blob(desc.data(), {(long)desc.size()}, torch::kInt64)
+                    .clone().to(x.device());
+  hipLaunchKernelGGL(k_seg_gram, dim3((unsigned)(desc.size() / 3)), dim3(NTHREADS), 0,
+                     cur_stream(), x.data_ptr<float>(), desc_t.data_ptr<long>(), (int)P,
+                     out.data_ptr<float>(), d);
+  return out;
+}
+
+// --------------------------------------------------------------------------- module
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_sgd_step", &fused_sgd_step);
+  m.def("fused_adam_step", &fused_adam_step);
+  m.def("inject", &inject);
+  m.def("rows_equal", &rows_equal);
+  m.def("mean_rows", &mean_rows);
+  m.def("sum_rows", &sum_rows);
+  m.def("cyclic_encode", &cyclic_encode);
+  m.def("cyclic_project", &cyclic_project);
+  m.def("cyclic_recombine", &cyclic_recombine);
+  m.def("segment_sqdist", &segment_sqdist);
+  m.def("segment_weighted_mean", &segment_weighted_mean);
+  m.def("segment_gram", &segment_gram);
+}

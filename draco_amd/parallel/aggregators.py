@@ -1,0 +1,246 @@
+"""Sharded robust-aggregation policies.
+
+Each policy turns the per-rank payload (the local logical workers' flat gradients)
+into the decoded global gradient, replicated on every rank.  All of them follow the
+same xGMI-friendly shape: one all_to_all of d/world shards, decode on the local shard
+(HIP kernels via draco_amd.ops), tiny allreduces for cross-shard decisions, one
+all_gather of the decoded shard.  Reference semantics per policy:
+
+  mean         baseline_master.py:202-207,267-269   (plain averaging)
+  maj_vote     rep_master.py:141-168                (group-wise Boyer-Moore vote)
+  geo_median   baseline_master.py:271-276           (per-layer Weiszfeld)
+  krum         baseline_master.py:278-296           (per-layer Krum selection)
+  cyclic       cyclic_master.py:103-188             (DFT-code algebraic decode)
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from .. import ops
+from ..coding import CyclicCode, majority_vote_index
+from .comm import Communicator
+from .flat import FlatSpace
+
+
+class Aggregator:
+    """Base: owns comm + space, provides the gather/allgather plumbing."""
+
+    name = "base"
+
+    def __init__(self, comm: Communicator, space: FlatSpace):
+        self.comm = comm
+        self.space = space
+        self._out = torch.zeros(space.d_pad, dtype=torch.float32, device=space.device)
+        self._shard_out = torch.zeros(space.shard, dtype=torch.float32, device=space.device)
+        self.local_seg = space.local_seg_bounds(comm.rank).to(space.device)
+
+    def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
+        raise NotImplementedError
+
+
+class MeanAggregator(Aggregator):
+    """Plain averaging over all P = L*world logical workers (reduce-scatter path:
+    no all_to_all needed since mean commutes with sharding)."""
+
+    name = "mean"
+
+    def __init__(self, comm, space, num_workers: int):
+        super().__init__(comm, space)
+        self.num_workers = num_workers
+
+    def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
+        local_sum = self._out  # reuse as scratch for the local sum
+        ops.sum_rows(payload, local_sum)
+        shard_sum = self.comm.reduce_scatter_sum(local_sum)
+        shard_sum /= float(self.num_workers)
+        self.comm.all_gather_shard(shard_sum, self._out)
+        return self._out
+
+
+class VoteAggregator(Aggregator):
+    """Repetition-code decode: per-group majority vote, then mean of winners.
+
+    Colocated layout: G = world groups of size r; member i of group g is local worker
+    slot i of rank (g+i) % world, so received row of member (g, i) is
+    src*L + l with src = (g+i) % world, l = i.
+    """
+
+    name = "maj_vote"
+
+    def __init__(self, comm, space, group_size: int, atol: float = 0.0, member_rows=None):
+        super().__init__(comm, space)
+        self.atol = atol
+        if member_rows is None:
+            # colocated layout: G = world groups; member i of group g is row
+            # ((g+i)%world)*r + i of the all_to_all result
+            r = group_size
+            member_rows = np.asarray(
+                [[((g + i) % comm.world) * r + i for i in range(r)] for g in range(max(comm.world, 1))]
+            )
+        self.member_rows = np.asarray(member_rows)  # (G, r)
+        self.G, self.r = self.member_rows.shape
+        pairs_a, pairs_b = [], []
+        for rows in self.member_rows:
+            for i in range(self.r):
+                for j in range(i + 1, self.r):
+                    pairs_a.append(rows[i])
+                    pairs_b.append(rows[j])
+        self.pairs_a = torch.tensor(pairs_a, dtype=torch.int64, device=space.device)
+        self.pairs_b = torch.tensor(pairs_b, dtype=torch.int64, device=space.device)
+        self.n_pairs_per_group = self.r * (self.r - 1) // 2
+
+    @classmethod
+    def from_member_rows(cls, comm, space, member_rows, atol: float = 0.0):
+        return cls(comm, space, group_size=member_rows.shape[1], atol=atol, member_rows=member_rows)
+
+    def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
+        recv = self.comm.all_to_all_rows(payload)  # (world*r, shard)
+        eq = ops.rows_equal(recv, self.pairs_a, self.pairs_b, self.atol)
+        self.comm.all_reduce(eq, op="min")  # equal iff equal on every shard
+        eq_host = eq.to("cpu", non_blocking=False).numpy()
+        winners = np.empty(self.G, dtype=np.int64)
+        k = 0
+        for g in range(self.G):
+            mat = np.eye(self.r, dtype=bool)
+            for i in range(self.r):
+                for j in range(i + 1, self.r):
+                    mat[i, j] = mat[j, i] = bool(eq_host[k])
+                    k += 1
+            winners[g] = self.member_rows[g, majority_vote_index(mat)]
+        idx = torch.tensor(winners, dtype=torch.int64, device=recv.device)
+        ops.mean_rows(recv, idx, self._shard_out)
+        self.comm.all_gather_shard(self._shard_out, self._out)
+        return self._out
+
+
+class GeoMedianAggregator(Aggregator):
+    """Per-layer geometric median (Weiszfeld), sharded over d.
+
+    Each iteration: per-(worker, layer) partial squared distances on the local shard,
+    one (P, L) allreduce, then the reweighted mean on the shard.  Every rank sees the
+    identical allreduced distances, so the iterates stay consistent without any
+    further synchronisation.
+    """
+
+    name = "geo_median"
+
+    def __init__(self, comm, space, num_workers: int, max_iter: int = 80, tol: float = 1e-7):
+        super().__init__(comm, space)
+        self.num_workers = num_workers
+        self.max_iter = max_iter
+        self.tol = tol
+
+    def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
+        recv = self.comm.all_to_all_rows(payload)  # (P, shard)
+        P = recv.shape[0]
+        z = recv.mean(dim=0)  # init at the mean (hdmedians does the same)
+        z_new = torch.empty_like(z)
+        for _ in range(self.max_iter):
+            part = ops.segment_sqdist(recv, z, self.local_seg)  # (P, L)
+            self.comm.all_reduce(part)
+            dist = part.clamp_min(1e-24).sqrt()
+            w = 1.0 / dist
+            w = w / w.sum(dim=0, keepdim=True)
+            ops.segment_weighted_mean(recv, w, self.local_seg, z_new)
+            delta = (z_new - z).pow(2).sum()
+            self.comm.all_reduce(delta)
+            scale = z_new.pow(2).sum()
+            self.comm.all_reduce(scale)
+            z, z_new = z_new, z
+            if float(delta) <= self.tol * self.tol * max(float(scale), 1e-12):
+                break
+        self.comm.all_gather_shard(z, self._out)
+        return self._out
+
+
+class KrumAggregator(Aggregator):
+    """Per-layer Krum selection (arXiv:1703.02757), sharded over d.
+
+    Pairwise squared distances come from per-segment Gram matrices (one allreduce of
+    (L, P, P)); selection score per layer = sum of the P-s-2 smallest distances to the
+    other workers, argmin wins (exactly baseline_master.py:279-291).
+    """
+
+    name = "krum"
+
+    def __init__(self, comm, space, num_workers: int, s: int):
+        super().__init__(comm, space)
+        self.num_workers = num_workers
+        self.s = s
+
+    def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
+        recv = self.comm.all_to_all_rows(payload)  # (P, shard)
+        P = recv.shape[0]
+        gram = ops.segment_gram(recv, self.local_seg)  # (L, P, P)
+        self.comm.all_reduce(gram)
+        g = gram.to("cpu").numpy().astype(np.float64)
+        diag = np.einsum("lpp->lp", g)
+        d2 = diag[:, :, None] + diag[:, None, :] - 2.0 * g  # (L, P, P)
+        np.maximum(d2, 0.0, out=d2)
+        L = d2.shape[0]
+        keep = max(self.num_workers - self.s - 2, 1)
+        winners = np.empty(L, dtype=np.int64)
+        for l in range(L):
+            m = d2[l].copy()
+            np.fill_diagonal(m, np.inf)
+            m.sort(axis=1)
+            scores = m[:, :keep].sum(axis=1)
+            winners[l] = int(np.argmin(scores))
+        # build the output shard layer-by-layer from each layer's winning worker
+        seg = self.local_seg.to("cpu").numpy()
+        out = self._shard_out
+        for l in range(L):
+            lo, hi = int(seg[l]), int(seg[l + 1])
+            if hi > lo:
+                out[lo:hi] = recv[winners[l], lo:hi]
+        tail = int(seg[-1])
+        if tail < out.shape[0]:
+            out[tail:] = 0.0
+        self.comm.all_gather_shard(out, self._out)
+        return self._out
+
+
+class CyclicAggregator(Aggregator):
+    """Cyclic-code algebraic decode, sharded over d.
+
+    The payload rows here are the ENCODED complex gradients as (2, d) fp32 planes —
+    (L*2, d_pad) per rank.  Decode: random projection proj = R @ z (partial per shard,
+    one (n, 2) allreduce), host error-location + recombination-vector solve (tiny,
+    cached per healthy-set), then out = Re(v @ R)/n on the shard.
+    """
+
+    name = "cyclic"
+
+    def __init__(self, comm, space, code: CyclicCode, workers_per_rank: int):
+        super().__init__(comm, space)
+        self.code = code
+        self.L = workers_per_rank
+        self.n = code.n
+
+    def aggregate(self, payload_planes: torch.Tensor, step: int) -> torch.Tensor:
+        # payload_planes: (L*2, d_pad); worker w = rank*L + l owns rows (2l, 2l+1)
+        recv = self.comm.all_to_all_rows(payload_planes)  # (world*L*2, shard)
+        r_planes = recv.view(self.n, 2, self.space.shard)
+        gen = torch.Generator(device="cpu")
+        gen.manual_seed(0x5EED ^ (step * 1000003) ^ self.comm.rank)
+        z = torch.normal(
+            mean=1.0, std=1.0, size=(self.space.shard,), generator=gen, dtype=torch.float32
+        ).to(self.space.device)
+        proj = ops.cyclic_project(r_planes, z)  # (n, 2) partial
+        self.comm.all_reduce(proj)
+        proj_c = proj.to("cpu").numpy().astype(np.float64)
+        proj_complex = proj_c[:, 0] + 1j * proj_c[:, 1]
+        syndrome = self.code.W_perp @ proj_complex
+        scale = float(np.abs(proj_complex).max())
+        if float(np.abs(syndrome).max()) <= 1e-7 * max(scale, 1e-30):
+            healthy = np.arange(self.n)
+        else:
+            healthy = self.code.locate_errors(syndrome)
+        v = self.code.recombination_vector(healthy)
+        v_re = torch.tensor(np.real(v), dtype=torch.float32)
+        v_im = torch.tensor(np.imag(v), dtype=torch.float32)
+        ops.cyclic_recombine(r_planes, v_re.to(recv.device), v_im.to(recv.device), self._shard_out)
+        self._shard_out /= float(self.n)
+        self.comm.all_gather_shard(self._shard_out, self._out)
+        return self._out

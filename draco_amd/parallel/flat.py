@@ -1,0 +1,92 @@
+"""Flat parameter / gradient space.
+
+The reference transmits and aggregates gradients tensor-by-tensor (62 MPI messages per
+worker per step for ResNet-18, baseline_master.py:188-200).  On MI355X with 288 GB of
+HBM3E the idiomatic layout is ONE contiguous fp32 buffer for all parameters and one
+(L, d) payload buffer for the L local logical workers' gradients:
+
+  * p.data is re-pointed at a view of the flat param buffer (zero-copy updates by the
+    fused SGD/Adam kernel over one contiguous range);
+  * before each logical worker's backward, p.grad is re-pointed at views of that
+    worker's payload row, so autograd accumulates gradients directly into the comm
+    buffer — the GPU analog of the reference's interleaved per-layer isends, with no
+    packing pass at all;
+  * d is padded so every rank owns an aligned d/world shard (all_to_all / reduce
+    scatter friendly) — shard alignment 64 floats = 256 B.
+
+Layer segment metadata (offsets per parameter tensor) is kept so the per-layer robust
+aggregation semantics of the reference (geo-median / Krum per parameter tensor,
+baseline_master.py:267-296) are preserved over the flat buffer.
+"""
+from __future__ import annotations
+
+import torch
+
+ALIGN = 64  # floats; 256 B
+
+
+def _pad(d: int, world: int) -> int:
+    q = world * ALIGN
+    return (d + q - 1) // q * q
+
+
+class FlatSpace:
+    def __init__(self, model: torch.nn.Module, world: int, device: torch.device):
+        self.device = device
+        self.world = world
+        params = [p for p in model.parameters() if p.requires_grad]
+        self.params = params
+        self.shapes = [p.shape for p in params]
+        self.numels = [p.numel() for p in params]
+        self.offsets = []
+        off = 0
+        for n in self.numels:
+            self.offsets.append(off)
+            off += n
+        self.d = off
+        self.d_pad = _pad(off, world)
+        self.shard = self.d_pad // world
+
+        self.flat_param = torch.zeros(self.d_pad, dtype=torch.float32, device=device)
+        with torch.no_grad():
+            for p, o, n in zip(params, self.offsets, self.numels):
+                self.flat_param[o : o + n].copy_(p.data.reshape(-1).to(device=device, dtype=torch.float32))
+                p.data = self.flat_param[o : o + n].view(p.shape)
+
+        # (L+1) segment bounds in global flat coordinates (one segment per parameter)
+        self.seg_bounds = torch.tensor(self.offsets + [self.d], dtype=torch.int64)
+
+    # ------------------------------------------------------------------ grads
+    def alloc_payload(self, rows: int) -> torch.Tensor:
+        return torch.zeros(rows, self.d_pad, dtype=torch.float32, device=self.device)
+
+    def attach_grads(self, buf: torch.Tensor) -> None:
+        """Point every parameter's .grad at views of the given flat (d_pad,) buffer."""
+        assert buf.shape == (self.d_pad,)
+        for p, o, n in zip(self.params, self.offsets, self.numels):
+            p.grad = buf[o : o + n].view(p.shape)
+
+    def detach_grads(self) -> None:
+        for p in self.params:
+            p.grad = None
+
+    # ------------------------------------------------------------------ shards
+    def local_seg_bounds(self, rank: int) -> torch.Tensor:
+        """Segment bounds clipped to this rank's shard, in shard-local coordinates.
+
+        Returns (L+1,) int64 monotone bounds (segments outside the shard are empty).
+        """
+        lo = rank * self.shard
+        hi = lo + self.shard
+        return (self.seg_bounds.clamp(min=lo, max=hi) - lo).contiguous()
+
+    def shard_of(self, flat: torch.Tensor, rank: int) -> torch.Tensor:
+        return flat[rank * self.shard : (rank + 1) * self.shard]
+
+    # ------------------------------------------------------------------ io
+    def state_dict_tensors(self):
+        return {"flat_param": self.flat_param[: self.d].clone()}
+
+    def load_flat(self, flat: torch.Tensor):
+        with torch.no_grad():
+            self.flat_param[: self.d].copy_(flat.to(self.device))

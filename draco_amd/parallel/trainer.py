@@ -1,0 +1,254 @@
+"""Colocated trainer: every rank computes gradients AND participates in the sharded
+robust decode.  This replaces the reference's dedicated parameter-server process
+(/root/reference/src/master/*, worker/*) — on an 8-GPU xGMI node a pure PS GPU would
+idle 1/8 of the compute and bottleneck all gradient traffic on one GPU's links, so
+the decode is sharded across all ranks instead (parallel/aggregators.py).  A
+reference-parity PS topology lives in parallel/ps.py.
+
+Logical-worker layout per approach (P = total logical workers):
+  baseline   L=1 per rank, P=world; every worker draws its own batch.
+  maj_vote   L=r per rank, G=world groups; member i of group g is slot i of rank
+             (g+i)%world, so group members sit on different GPUs (the vote actually
+             crosses the wire) and per-GPU work is r forward/backward per step.
+  cyclic     L=n/world workers per rank (n=P); each computes the 2s+1 sub-batch
+             gradients of its cyclic band and ships one encoded complex gradient.
+"""
+from __future__ import annotations
+
+import math
+import os
+import time
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+from ..coding import AdversarySchedule, build_cyclic_code
+from ..config import Config
+from ..data import GlobalBatchSource, GroupBatchSource, SyntheticClassification
+from ..models import build_model
+from ..optim import FlatSGD
+from ..utils.checkpoint import load_checkpoint, save_checkpoint
+from ..utils.logging import MetricsLogger
+from .aggregators import (
+    CyclicAggregator,
+    GeoMedianAggregator,
+    KrumAggregator,
+    MeanAggregator,
+    VoteAggregator,
+)
+from .comm import Communicator
+from .flat import FlatSpace
+
+
+def _resolve_device(cfg: Config, rank: int) -> torch.device:
+    if cfg.device == "cpu":
+        return torch.device("cpu")
+    if cfg.device == "cuda" or (cfg.device == "auto" and torch.cuda.is_available()):
+        local = int(os.environ.get("LOCAL_RANK", rank % max(torch.cuda.device_count(), 1)))
+        torch.cuda.set_device(local)
+        return torch.device("cuda", local)
+    return torch.device("cpu")
+
+
+class Trainer:
+    def __init__(self, cfg: Config, comm: Communicator | None = None):
+        cfg.sanity()
+        self.cfg = cfg
+        rank = int(os.environ.get("RANK", 0))
+        device = _resolve_device(cfg, rank)
+        self.comm = comm or Communicator.from_env(device)
+        self.device = device
+        self.rank = self.comm.rank
+        self.world = self.comm.world
+
+        if cfg.deterministic:
+            torch.backends.cudnn.deterministic = True
+            torch.backends.cudnn.benchmark = False
+        else:
+            torch.backends.cudnn.benchmark = True
+
+        # identical init on every rank
+        torch.manual_seed(cfg.seed)
+        self.model = build_model(cfg.network, cfg.dataset).to(device)
+        self.model.train()
+        self.space = FlatSpace(self.model, self.world, device)
+        self.opt = FlatSGD(self.space.flat_param, lr=cfg.lr, momentum=cfg.momentum)
+
+        self.autocast_dtype = torch.bfloat16 if (cfg.dtype == "bf16" and device.type == "cuda") else None
+
+        # ---------------- logical workers & aggregator ----------------
+        approach = cfg.approach
+        self.approach = approach
+        if approach == "baseline":
+            self.L = 1
+            self.P = self.world
+            self.agg = self._baseline_aggregator(cfg)
+            self.data = GroupBatchSource(self._dataset(), cfg.batch_size, n_groups=self.P)
+            self.payload = self.space.alloc_payload(self.L)
+        elif approach == "maj_vote":
+            self.r = cfg.group_size
+            self.L = self.r
+            self.G = self.world
+            self.P = self.L * self.world
+            self.agg = VoteAggregator(self.comm, self.space, group_size=self.r, atol=cfg.vote_atol)
+            self.data = GroupBatchSource(self._dataset(), cfg.batch_size, n_groups=self.G)
+            self.payload = self.space.alloc_payload(self.L)
+        elif approach == "cyclic":
+            wpr = cfg.workers_per_rank or max(1, math.ceil((2 * cfg.worker_fail + 2) / self.world))
+            self.L = wpr
+            self.n = self.L * self.world
+            self.P = self.n
+            self.code = build_cyclic_code(self.n, cfg.worker_fail)
+            self.s_hat = self.code.s_hat
+            self.agg = CyclicAggregator(self.comm, self.space, self.code, self.L)
+            self.data = GlobalBatchSource(self._dataset(), cfg.batch_size, n_workers=self.n)
+            # encoded complex payload planes + raw sub-batch gradient scratch
+            self.payload = self.space.alloc_payload(self.L * 2)
+            self.scratch = self.space.alloc_payload(self.s_hat)
+            W = self.code.W
+            sup = self.code.support
+            self._w_re, self._w_im = [], []
+            for l in range(self.L):
+                w_global = self.rank * self.L + l
+                coeff = W[w_global, sup[w_global]]
+                self._w_re.append(torch.tensor(np.real(coeff), dtype=torch.float32, device=device))
+                self._w_im.append(torch.tensor(np.imag(coeff), dtype=torch.float32, device=device))
+        else:
+            raise ValueError(f"unknown approach {approach!r}")
+
+        self.n_fail = min(cfg.worker_fail, self.P)
+        self.schedule = AdversarySchedule(self.P, self.n_fail, cfg.max_steps)
+        self.step_num = 0
+        self.logger = MetricsLogger(cfg.log_dir, self.rank)
+        self.criterion = F.cross_entropy
+
+        if cfg.checkpoint_step > 0:
+            self.load(cfg.checkpoint_step)
+
+    # ------------------------------------------------------------------ helpers
+    def _dataset(self) -> SyntheticClassification:
+        return SyntheticClassification(self.cfg.dataset, self.device, seed=1234)
+
+    def _baseline_aggregator(self, cfg: Config):
+        if cfg.mode == "normal":
+            return MeanAggregator(self.comm, self.space, num_workers=self.world)
+        if cfg.mode == "geometric_median":
+            return GeoMedianAggregator(self.comm, self.space, num_workers=self.world)
+        if cfg.mode == "krum":
+            return KrumAggregator(self.comm, self.space, num_workers=self.world, s=cfg.worker_fail)
+        raise ValueError(f"baseline approach supports modes normal/geometric_median/krum, got {cfg.mode!r}")
+
+    def _forward_backward(self, x, y, grad_row: torch.Tensor) -> float:
+        self.space.attach_grads(grad_row)
+        grad_row.zero_()
+        if self.autocast_dtype is not None:
+            with torch.autocast("cuda", dtype=self.autocast_dtype):
+                logits = self.model(x)
+                loss = self.criterion(logits, y)
+        else:
+            logits = self.model(x)
+            loss = self.criterion(logits, y)
+        loss.backward()
+        return float(loss.detach())
+
+    # ------------------------------------------------------------------ one step
+    def train_step(self) -> dict:
+        cfg = self.cfg
+        step = self.step_num
+        t0 = time.perf_counter()
+        losses = []
+        adversaries = self.schedule.adversaries_at(step) if self.n_fail > 0 else frozenset()
+
+        if self.approach in ("baseline", "maj_vote"):
+            for l in range(self.L):
+                if self.approach == "baseline":
+                    group = self.rank  # every worker draws its own stream
+                    worker_id = self.rank
+                else:
+                    group = (self.rank - l) % self.world
+                    worker_id = ((group + l) % self.world) * self.L + l  # == rank*L + l
+                x, y = self.data.batch_for(group, step)
+                losses.append(self._forward_backward(x, y, self.payload[l]))
+                if worker_id in adversaries:
+                    ops.inject_(self.payload[l], cfg.err_mode, cyclic=False)
+        else:  # cyclic
+            for l in range(self.L):
+                w_global = self.rank * self.L + l
+                sup = self.code.support[w_global]
+                for k in range(self.s_hat):
+                    x, y = self.data.sub_batch(int(sup[k]), step)
+                    losses.append(self._forward_backward(x, y, self.scratch[k]))
+                enc = self.payload[2 * l : 2 * l + 2]
+                ops.cyclic_encode(self.scratch, self._w_re[l], self._w_im[l], enc)
+                if w_global in adversaries:
+                    self._inject_encoded(enc, cfg.err_mode)
+
+        t_comp = time.perf_counter()
+        grad = self.agg.aggregate(self.payload, step)
+        t_agg = time.perf_counter()
+        self.opt.step(grad)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+
+        self.step_num += 1
+        rec = {
+            "step": step,
+            "loss": float(np.mean(losses)),
+            "time": t1 - t0,
+            "comp": t_comp - t0,
+            "agg": t_agg - t_comp,
+            "update": t1 - t_agg,
+        }
+        if cfg.eval_freq > 0 and self.step_num % cfg.eval_freq == 0:
+            if self.rank == 0:
+                self.save()
+        self.logger.log(rec)
+        return rec
+
+    def _inject_encoded(self, enc: torch.Tensor, mode: str) -> None:
+        # reference injects on the encoded complex combination with cyclic=True
+        # (additive error, cyclic_worker.py:181-183 / model_ops/utils.py:8-11)
+        if mode == "rev_grad":
+            enc.add_(enc, alpha=ops.fallback.ADVERSARY_)
+        elif mode == "constant":
+            enc[0].add_(ops.fallback.ADVERSARY_)
+        elif mode in ("random", "none", ""):
+            pass
+        elif mode == "gauss":
+            enc.add_(torch.randn_like(enc) * enc.abs().mean().clamp(min=1e-12) * 100.0)
+        else:
+            raise ValueError(mode)
+
+    # ------------------------------------------------------------------ eval / io
+    @torch.no_grad()
+    def evaluate(self, n_batches: int = 8) -> dict:
+        self.model.eval()
+        data = self._dataset()
+        correct = total = 0
+        loss_sum = 0.0
+        for b in range(n_batches):
+            # held-out stream: negative index space never touched by training
+            x, y = data.get_batch(2**40 + b * self.cfg.test_batch_size, self.cfg.test_batch_size)
+            logits = self.model(x)
+            loss_sum += float(self.criterion(logits, y))
+            correct += int((logits.argmax(dim=1) == y).sum())
+            total += y.numel()
+        self.model.train()
+        return {"prec1": correct / total, "loss": loss_sum / n_batches}
+
+    def _ckpt_path(self, step: int | None = None) -> str:
+        step = self.step_num if step is None else step
+        return os.path.join(self.cfg.train_dir, f"model_step_{step}")
+
+    def save(self):
+        save_checkpoint(self._ckpt_path(), self.model, self.space, self.opt, self.step_num, self.cfg)
+
+    def load(self, step: int):
+        load_checkpoint(self._ckpt_path(step), self.model, self.space, self.opt)
+        self.step_num = step
+
+    def close(self):
+        self.logger.close()

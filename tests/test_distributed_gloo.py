@@ -1,0 +1,191 @@
+"""Multi-process (gloo, CPU) tests of the sharded aggregation paths — the distributed
+correctness lane that runs without GPUs (world_size 2-4)."""
+import numpy as np
+import pytest
+import torch
+
+from tests.dist_util import run_dist
+
+
+def _comm(rank, world):
+    from draco_amd.parallel.comm import Communicator
+
+    return Communicator(rank, world, torch.device("cpu"), backend="gloo")
+
+
+# --------------------------------------------------------------------- all_to_all
+def _a2a_worker(rank, world):
+    comm = _comm(rank, world)
+    L, shard = 2, 8
+    d_pad = world * shard
+    payload = torch.zeros(L, d_pad)
+    for l in range(L):
+        payload[l] = rank * 100 + l * 10 + torch.arange(d_pad, dtype=torch.float32) / 100.0
+    recv = comm.all_to_all_rows(payload)
+    # row w = src*L + l must hold src's worker-l shard for this rank
+    for src in range(world):
+        for l in range(L):
+            expect = src * 100 + l * 10 + (torch.arange(shard) + rank * shard) / 100.0
+            assert torch.allclose(recv[src * L + l], expect.float()), (rank, src, l)
+    comm.shutdown()
+    return True
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_all_to_all_rows(world):
+    run_dist(_a2a_worker, world)
+
+
+# --------------------------------------------------------------------- trainer e2e
+def _trainer_worker(rank, world, approach, mode, kw):
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                 approach=approach, mode=mode, max_steps=50, eval_freq=0, log_dir="",
+                 train_dir="/tmp/draco_test_ckpt", **kw)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    losses = [t.train_step()["loss"] for _ in range(8)]
+    # flat params must remain bit-identical across ranks (identical decoded updates)
+    param_hash = float(t.space.flat_param.double().sum())
+    t.close()
+    t.comm.shutdown()
+    return (losses, param_hash)
+
+
+@pytest.mark.parametrize(
+    "world,approach,mode,kw",
+    [
+        (2, "baseline", "normal", dict(worker_fail=0)),
+        (3, "maj_vote", "maj_vote", dict(group_size=3, worker_fail=1)),
+        (2, "maj_vote", "maj_vote", dict(group_size=3, worker_fail=1)),
+        (2, "cyclic", "cyclic", dict(worker_fail=1, workers_per_rank=2)),
+        (2, "baseline", "geometric_median", dict(worker_fail=0)),
+        (2, "baseline", "krum", dict(worker_fail=0)),
+    ],
+)
+def test_trainer_distributed(world, approach, mode, kw):
+    res = run_dist(_trainer_worker, world, approach, mode, kw)
+    hashes = [res[r][1] for r in range(world)]
+    assert all(h == hashes[0] for h in hashes), "replicated params diverged across ranks"
+    losses0 = res[0][0]
+    assert losses0[-1] < losses0[0]
+
+
+# ----------------------------------------------------- distributed == local decode
+def _vote_equiv_worker(rank, world):
+    """Sharded vote over 3 ranks must reproduce the single-process (unsharded) vote."""
+    from draco_amd.parallel.aggregators import VoteAggregator
+    from draco_amd.parallel.comm import Communicator
+    from draco_amd.parallel.flat import FlatSpace
+    import torch.nn as nn
+
+    comm = _comm(rank, world)
+    torch.manual_seed(7)
+    model = nn.Linear(50, 10)
+    space = FlatSpace(model, world, torch.device("cpu"))
+    r = 3
+    agg = VoteAggregator(comm, space, group_size=r, atol=0.0)
+    # deterministic worker gradients: member i of group g has gradient f(g) except
+    # one adversarial member per group
+    payload = space.alloc_payload(r)
+    for l in range(r):
+        g = (rank - l) % world
+        torch.manual_seed(1000 + g)
+        honest = torch.randn(space.d_pad)
+        member_rank = rank  # member l of group g lives here
+        # adversary: member (g % r) of each group
+        if l == g % r:
+            payload[l] = honest * -100.0
+        else:
+            payload[l] = honest
+    out = agg.aggregate(payload, step=0)
+    # expected: mean over groups of honest gradients
+    ref = torch.zeros(space.d_pad)
+    for g in range(world):
+        torch.manual_seed(1000 + g)
+        ref += torch.randn(space.d_pad)
+    ref /= world
+    assert torch.allclose(out, ref, atol=1e-6), float((out - ref).abs().max())
+    comm.shutdown()
+    return True
+
+
+def test_sharded_vote_equals_local():
+    run_dist(_vote_equiv_worker, 3)
+
+
+def _mean_equiv_worker(rank, world):
+    from draco_amd.parallel.aggregators import MeanAggregator
+    from draco_amd.parallel.comm import Communicator
+    from draco_amd.parallel.flat import FlatSpace
+    import torch.nn as nn
+
+    comm = _comm(rank, world)
+    torch.manual_seed(7)
+    model = nn.Linear(40, 4)
+    space = FlatSpace(model, world, torch.device("cpu"))
+    agg = MeanAggregator(comm, space, num_workers=world)
+    payload = space.alloc_payload(1)
+    torch.manual_seed(50 + rank)
+    payload[0] = torch.randn(space.d_pad)
+    out = agg.aggregate(payload, step=0)
+    ref = torch.zeros(space.d_pad)
+    for rr in range(world):
+        torch.manual_seed(50 + rr)
+        ref += torch.randn(space.d_pad)
+    ref /= world
+    assert torch.allclose(out, ref, atol=1e-6)
+    comm.shutdown()
+    return True
+
+
+def test_sharded_mean_equals_local():
+    run_dist(_mean_equiv_worker, 2)
+
+
+def _cyclic_equiv_worker(rank, world):
+    """Sharded cyclic decode must recover the exact sum of sub-batch gradients with
+    one corrupted worker."""
+    import torch.nn as nn
+
+    from draco_amd import ops
+    from draco_amd.coding import build_cyclic_code
+    from draco_amd.parallel.aggregators import CyclicAggregator
+    from draco_amd.parallel.flat import FlatSpace
+
+    comm = _comm(rank, world)
+    torch.manual_seed(7)
+    model = nn.Linear(64, 8)
+    space = FlatSpace(model, world, torch.device("cpu"))
+    L = 2
+    n = L * world
+    s = 1
+    code = build_cyclic_code(n, s)
+    agg = CyclicAggregator(comm, space, code, workers_per_rank=L)
+    # deterministic sub-batch "gradients"
+    def sub_grad(j):
+        torch.manual_seed(300 + j)
+        return torch.randn(space.d_pad)
+
+    payload = space.alloc_payload(L * 2)
+    for l in range(L):
+        w = rank * L + l
+        sup = code.support[w]
+        grads = torch.stack([sub_grad(int(j)) for j in sup])
+        wre = torch.tensor(np.real(code.W[w, sup]), dtype=torch.float32)
+        wim = torch.tensor(np.imag(code.W[w, sup]), dtype=torch.float32)
+        ops.cyclic_encode(grads, wre, wim, payload[2 * l : 2 * l + 2])
+        if w == 1:  # adversary
+            enc = payload[2 * l : 2 * l + 2]
+            enc.add_(enc, alpha=-100.0)
+    out = agg.aggregate(payload, step=3)
+    ref = torch.stack([sub_grad(j) for j in range(n)]).sum(0) / n
+    assert torch.allclose(out, ref, atol=1e-4), float((out - ref).abs().max())
+    comm.shutdown()
+    return True
+
+
+def test_sharded_cyclic_equals_local():
+    run_dist(_cyclic_equiv_worker, 2)
