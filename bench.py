@@ -62,7 +62,6 @@ def main():
         log_dir="",
         train_dir="gpurun_out/bench_ckpt",
         topology="colocated",
-        deterministic=True,
     )
     t = Trainer(cfg)
     t.logger.stdout_every = 0

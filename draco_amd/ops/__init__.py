@@ -15,6 +15,8 @@ __all__ = [
     "fused_adam_step",
     "inject_",
     "rows_equal",
+    "pair_maxdiff",
+    "row_absmax",
     "mean_rows",
     "sum_rows",
     "cyclic_encode",
@@ -81,6 +83,20 @@ def rows_equal(x, a_idx, b_idx, atol):
     if ext is not None:
         return ext.rows_equal(x, a_idx.to(x.device), b_idx.to(x.device), float(atol))
     return fallback.rows_equal(x, a_idx, b_idx, atol)
+
+
+def pair_maxdiff(x, a_idx, b_idx):
+    ext = _native_for(x)
+    if ext is not None:
+        return ext.pair_maxdiff(x, a_idx.to(x.device), b_idx.to(x.device))
+    return fallback.pair_maxdiff(x, a_idx, b_idx)
+
+
+def row_absmax(x):
+    ext = _native_for(x)
+    if ext is not None:
+        return ext.row_absmax(x)
+    return fallback.row_absmax(x)
 
 
 def mean_rows(x, idx, out):

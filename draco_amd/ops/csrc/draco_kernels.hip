@@ -226,6 +226,70 @@ torch::Tensor rows_equal(torch::Tensor x, torch::Tensor a_idx, torch::Tensor b_i
   return out;
 }
 
+// Tolerance-vote statistics: per-pair max|a-b| and per-row max|x| over the shard.
+// (MIOpen conv backward is not bitwise-reproducible on this stack, so the GPU vote is
+// tolerance-based: equal iff max|a-b| <= atol + rtol*max(rowmax_a, rowmax_b) — the
+// thresholds are combined host-side after a cross-rank MAX allreduce.)
+__device__ inline void atomic_max_f32(float *addr, float val) {
+  // valid for non-negative floats: IEEE ordering matches int ordering
+  atomicMax((int *)addr, __float_as_int(val));
+}
+
+template <bool PAIR>
+__global__ void k_absmax(const float4 *__restrict__ x, const long *__restrict__ ai,
+                         const long *__restrict__ bi, float *__restrict__ out, long d4,
+                         long stride4) {
+  int row = blockIdx.y;
+  const float4 *ra = x + (PAIR ? ai[row] : (long)row) * stride4;
+  const float4 *rb = PAIR ? (x + bi[row] * stride4) : nullptr;
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  float mx = 0.f;
+  for (; i < d4; i += stride) {
+    float4 a = ra[i];
+    if (PAIR) {
+      float4 b = rb[i];
+      a = make_float4(a.x - b.x, a.y - b.y, a.z - b.z, a.w - b.w);
+    }
+    mx = fmaxf(mx, fmaxf(fmaxf(fabsf(a.x), fabsf(a.y)), fmaxf(fabsf(a.z), fabsf(a.w))));
+  }
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_down(mx, off, WAVE));
+  __shared__ float wmax[NTHREADS / WAVE];
+  int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) wmax[wid] = mx;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float tot = wmax[0];
+    for (int wv = 1; wv < NTHREADS / WAVE; ++wv) tot = fmaxf(tot, wmax[wv]);
+    atomic_max_f32(&out[row], tot);
+  }
+}
+
+torch::Tensor pair_maxdiff(torch::Tensor x, torch::Tensor a_idx, torch::Tensor b_idx) {
+  CHECK_IN(x); CHECK_IN(a_idx); CHECK_IN(b_idx);
+  long k = a_idx.numel(), d = x.size(1);
+  auto out = torch::zeros({k}, torch::dtype(torch::kFloat32).device(x.device()));
+  if (d == 0 || k == 0) return out;
+  dim3 grid(n_blocks(d / 4 / 8, NTHREADS), (unsigned)k);
+  hipLaunchKernelGGL((k_absmax<true>), grid, dim3(NTHREADS), 0, cur_stream(),
+                     (const float4 *)x.data_ptr<float>(), a_idx.data_ptr<long>(),
+                     b_idx.data_ptr<long>(), out.data_ptr<float>(), d / 4, d / 4);
+  return out;
+}
+
+torch::Tensor row_absmax(torch::Tensor x) {
+  CHECK_IN(x);
+  long m = x.size(0), d = x.size(1);
+  auto out = torch::zeros({m}, torch::dtype(torch::kFloat32).device(x.device()));
+  if (d == 0 || m == 0) return out;
+  dim3 grid(n_blocks(d / 4 / 8, NTHREADS), (unsigned)m);
+  hipLaunchKernelGGL((k_absmax<false>), grid, dim3(NTHREADS), 0, cur_stream(),
+                     (const float4 *)x.data_ptr<float>(), nullptr, nullptr,
+                     out.data_ptr<float>(), d / 4, d / 4);
+  return out;
+}
+
 // --------------------------------------------------------------------------- combine
 // Generic weighted row combination out[j] = sum_i w[i] * x[rows[i]][j].
 // Serves: mean of group winners (K8), sum_rows, cyclic encode per-plane and the final
@@ -508,6 +572,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam_step", &fused_adam_step);
   m.def("inject", &inject);
   m.def("rows_equal", &rows_equal);
+  m.def("pair_maxdiff", &pair_maxdiff);
+  m.def("row_absmax", &row_absmax);
   m.def("mean_rows", &mean_rows);
   m.def("sum_rows", &sum_rows);
   m.def("cyclic_encode", &cyclic_encode);

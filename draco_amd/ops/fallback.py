@@ -124,6 +124,22 @@ def rows_equal(x: torch.Tensor, a_idx: torch.Tensor, b_idx: torch.Tensor, atol: 
 
 
 @torch.no_grad()
+def pair_maxdiff(x: torch.Tensor, a_idx: torch.Tensor, b_idx: torch.Tensor) -> torch.Tensor:
+    """Per-pair max|x[a]-x[b]| over the local shard.  (k,) fp32."""
+    if x.shape[1] == 0:
+        return torch.zeros(len(a_idx), dtype=torch.float32, device=x.device)
+    return (x[a_idx] - x[b_idx]).abs().amax(dim=1).float()
+
+
+@torch.no_grad()
+def row_absmax(x: torch.Tensor) -> torch.Tensor:
+    """Per-row max|x| over the local shard.  (m,) fp32."""
+    if x.shape[1] == 0:
+        return torch.zeros(x.shape[0], dtype=torch.float32, device=x.device)
+    return x.abs().amax(dim=1).float()
+
+
+@torch.no_grad()
 def mean_rows(x: torch.Tensor, idx: torch.Tensor, out: torch.Tensor) -> None:
     """out = mean over selected rows of x.  x: (m, d), idx: (k,), out: (d,)."""
     torch.mean(x[idx], dim=0, out=out)

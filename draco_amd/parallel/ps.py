@@ -116,7 +116,11 @@ class Master(_PSBase):
             if self.P % r != 0:
                 raise ValueError(f"ps maj_vote needs P % group_size == 0 (P={self.P}, r={r})")
             member_rows = np.arange(self.P).reshape(self.P // r, r)
-            self.agg = VoteAggregator.from_member_rows(local, self.space, member_rows, atol=cfg.vote_atol)
+            rtol = cfg.vote_rtol
+            if rtol < 0:
+                rtol = 1e-3 if self.device.type == "cuda" else 0.0
+            self.agg = VoteAggregator.from_member_rows(local, self.space, member_rows,
+                                                       atol=cfg.vote_atol, rtol=rtol)
         elif cfg.approach == "cyclic":
             self.agg = CyclicAggregator(local, self.space, self.code, workers_per_rank=self.n)
         self.gather_buf = self.space.alloc_payload(self.P * self.payload_rows)

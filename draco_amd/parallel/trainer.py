@@ -92,7 +92,12 @@ class Trainer:
             self.L = self.r
             self.G = self.world
             self.P = self.L * self.world
-            self.agg = VoteAggregator(self.comm, self.space, group_size=self.r, atol=cfg.vote_atol)
+            rtol = cfg.vote_rtol
+            if rtol < 0:  # auto: bitwise on CPU, tolerance on GPU (see VoteAggregator doc)
+                rtol = 1e-3 if device.type == "cuda" else 0.0
+            self.vote_rtol = rtol
+            self.agg = VoteAggregator(self.comm, self.space, group_size=self.r,
+                                      atol=cfg.vote_atol, rtol=rtol)
             self.data = GroupBatchSource(self._dataset(), cfg.batch_size, n_groups=self.G)
             self.payload = self.space.alloc_payload(self.L)
         elif approach == "cyclic":

