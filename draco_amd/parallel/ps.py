@@ -117,8 +117,8 @@ class Master(_PSBase):
                 raise ValueError(f"ps maj_vote needs P % group_size == 0 (P={self.P}, r={r})")
             member_rows = np.arange(self.P).reshape(self.P // r, r)
             rtol = cfg.vote_rtol
-            if rtol < 0:
-                rtol = 1e-3 if self.device.type == "cuda" else 0.0
+            if rtol < 0:  # auto (no autocast in ps workers -> fp32 noise on GPU)
+                rtol = 1e-4 if self.device.type == "cuda" else 0.0
             self.agg = VoteAggregator.from_member_rows(local, self.space, member_rows,
                                                        atol=cfg.vote_atol, rtol=rtol)
         elif cfg.approach == "cyclic":
