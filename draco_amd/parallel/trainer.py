@@ -97,10 +97,10 @@ class Trainer:
                 if device.type != "cuda":
                     rtol = 0.0
                 elif cfg.dtype == "bf16":
-                    # bf16 autocast: replica noise sits at bf16 rounding (~0.4% of the
-                    # row max under algo/order variation); 5e-2 gives ~10x margin while
-                    # a rev_grad adversary is ~2000x outside the threshold
-                    rtol = 5e-2
+                    # bf16 autocast: replica noise measured at ~1.2% of the row max
+                    # under MIOpen algo/order variation (tools/diag_det.py); 0.1 gives
+                    # ~8x margin while a rev_grad adversary is ~1000x outside
+                    rtol = 1e-1
                 else:
                     rtol = 1e-4  # fp32 reorder noise
             self.vote_rtol = rtol
