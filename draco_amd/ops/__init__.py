@@ -20,6 +20,7 @@ __all__ = [
     "mean_rows",
     "sum_rows",
     "cyclic_encode",
+    "combine_rows",
     "cyclic_project",
     "cyclic_recombine",
     "segment_sqdist",
@@ -123,11 +124,19 @@ def cyclic_encode(grads, w_re, w_im, out):
     fallback.cyclic_encode(grads, w_re, w_im, out)
 
 
-def cyclic_project(r_planes, z):
-    ext = _native_for(r_planes)
+def cyclic_project(rows, z):
+    ext = _native_for(rows)
     if ext is not None:
-        return ext.cyclic_project(r_planes, z)
-    return fallback.cyclic_project(r_planes, z)
+        return ext.cyclic_project(rows, z)
+    return fallback.cyclic_project(rows, z)
+
+
+def combine_rows(x, rows, w, out):
+    ext = _native_for(x)
+    if ext is not None:
+        ext.combine_rows(x, rows.to(x.device), w.to(x.device), out)
+        return
+    fallback.combine_rows(x, rows, w, out)
 
 
 def cyclic_recombine(r_planes, v_re, v_im, out):

@@ -415,18 +415,24 @@ __global__ void k_project(const float4 *__restrict__ r, const float4 *__restrict
   }
 }
 
-torch::Tensor cyclic_project(torch::Tensor r_planes, torch::Tensor z) {
-  CHECK_IN(r_planes); CHECK_IN(z);
-  long n = r_planes.size(0);
-  long d = r_planes.size(2);
-  auto out = torch::zeros({n, 2}, torch::dtype(torch::kFloat32).device(r_planes.device()));
+torch::Tensor cyclic_project(torch::Tensor rows, torch::Tensor z) {
+  CHECK_IN(rows); CHECK_IN(z);
+  TORCH_CHECK(rows.dim() == 2, "cyclic_project: rows must be 2D");
+  long m = rows.size(0);
+  long d = rows.size(1);
+  auto out = torch::zeros({m}, torch::dtype(torch::kFloat32).device(rows.device()));
   if (d == 0) return out;
-  dim3 grid(n_blocks(d / 4 / 8, NTHREADS), (unsigned)(2 * n));
+  dim3 grid(n_blocks(d / 4 / 8, NTHREADS), (unsigned)m);
   hipLaunchKernelGGL(k_project, grid, dim3(NTHREADS), 0, cur_stream(),
-                     (const float4 *)r_planes.data_ptr<float>(),
+                     (const float4 *)rows.data_ptr<float>(),
                      (const float4 *)z.data_ptr<float>(), out.data_ptr<float>(), d / 4,
                      d / 4);
   return out;
+}
+
+void combine_rows(torch::Tensor x, torch::Tensor rows, torch::Tensor w, torch::Tensor out) {
+  CHECK_IN(x); CHECK_IN(rows); CHECK_IN(w); CHECK_IN(out);
+  launch_combine(x, rows, w, out, x.size(1));
 }
 
 // --------------------------------------------------------------------------- geomed
@@ -578,6 +584,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sum_rows", &sum_rows);
   m.def("cyclic_encode", &cyclic_encode);
   m.def("cyclic_project", &cyclic_project);
+  m.def("combine_rows", &combine_rows);
   m.def("cyclic_recombine", &cyclic_recombine);
   m.def("segment_sqdist", &segment_sqdist);
   m.def("segment_weighted_mean", &segment_weighted_mean);

@@ -163,13 +163,25 @@ def cyclic_encode(grads: torch.Tensor, w_re: torch.Tensor, w_im: torch.Tensor, o
 
 
 @torch.no_grad()
-def cyclic_project(r_planes: torch.Tensor, z: torch.Tensor) -> torch.Tensor:
-    """Partial projection proj_i = sum_d R[i,d] * z[d] over the local shard.
+def cyclic_project(rows: torch.Tensor, z: torch.Tensor) -> torch.Tensor:
+    """Partial per-row projection proj_r = sum_d rows[r, d] * z[d] over the shard.
 
-    r_planes: (n, 2, d_shard) fp32; z: (d_shard,) fp32 -> (n, 2) fp32 partials
+    rows: (m, d_shard) fp32; z: (d_shard,) fp32 -> (m,) fp32 partials
     (to be summed across ranks).
     """
-    return torch.einsum("npd,d->np", r_planes, z)
+    if rows.shape[1] == 0:
+        return torch.zeros(rows.shape[0], dtype=torch.float32, device=rows.device)
+    return rows @ z
+
+
+@torch.no_grad()
+def combine_rows(x: torch.Tensor, rows: torch.Tensor, w: torch.Tensor, out: torch.Tensor) -> None:
+    """out = sum_i w[i] * x[rows[i]] — the generic weighted row combination behind
+    winner-mean, cyclic encode/recombine and plain summing."""
+    if x.shape[1] == 0:
+        out.zero_()
+        return
+    torch.matmul(w, x[rows], out=out)
 
 
 @torch.no_grad()

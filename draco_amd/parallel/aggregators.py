@@ -24,7 +24,13 @@ from .flat import FlatSpace
 
 
 class Aggregator:
-    """Base: owns comm + space, provides the gather/allgather plumbing."""
+    """Base: owns comm + space, provides the exchange/allgather plumbing.
+
+    Overlap: the trainer may call start_row(payload, r) as soon as payload row r is
+    final (backward done + adversary injected) — the all_to_all for that row then
+    runs on the communication stream while the next logical worker's backward
+    computes.  aggregate() exchanges any not-yet-started rows and waits.
+    """
 
     name = "base"
 
@@ -34,6 +40,35 @@ class Aggregator:
         self._out = torch.zeros(space.d_pad, dtype=torch.float32, device=space.device)
         self._shard_out = torch.zeros(space.shard, dtype=torch.float32, device=space.device)
         self.local_seg = space.local_seg_bounds(comm.rank).to(space.device)
+        self._recv = None
+        self._works: list = []
+        self._started: set = set()
+
+    # ---------------------------------------------------------- row exchange
+    def start_row(self, payload: torch.Tensor, row: int) -> None:
+        if not self.comm.distributed:
+            return
+        if self._recv is None or self._recv.shape[0] != payload.shape[0]:
+            self._recv = torch.empty(payload.shape[0], self.comm.world, self.space.shard,
+                                     dtype=payload.dtype, device=payload.device)
+        w = self.comm.all_to_all_row(payload[row], self._recv[row], async_op=True)
+        if w is not None:
+            self._works.append(w)
+        self._started.add(row)
+
+    def exchanged(self, payload: torch.Tensor) -> torch.Tensor:
+        """(rows, d_pad) payload -> (rows*world, shard) in l-major row order."""
+        rows = payload.shape[0]
+        if not self.comm.distributed:
+            return payload.view(rows, self.space.shard)
+        for r in range(rows):
+            if r not in self._started:
+                self.start_row(payload, r)
+        for w in self._works:
+            w.wait()
+        self._works = []
+        self._started = set()
+        return self._recv.view(rows * self.comm.world, self.space.shard)
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
         raise NotImplementedError
@@ -86,11 +121,12 @@ class VoteAggregator(Aggregator):
         self.atol = atol
         self.rtol = rtol
         if member_rows is None:
-            # colocated layout: G = world groups; member i of group g is row
-            # ((g+i)%world)*r + i of the all_to_all result
-            r = group_size
+            # colocated layout: G = world groups; member i of group g is local slot i
+            # of rank (g+i)%world = row i*world + (g+i)%world (l-major convention of
+            # comm.all_to_all_rows)
             member_rows = np.asarray(
-                [[((g + i) % comm.world) * r + i for i in range(r)] for g in range(max(comm.world, 1))]
+                [[i * comm.world + (g + i) % comm.world for i in range(group_size)]
+                 for g in range(max(comm.world, 1))]
             )
         self.member_rows = np.asarray(member_rows)  # (G, r)
         self.G, self.r = self.member_rows.shape
@@ -110,7 +146,7 @@ class VoteAggregator(Aggregator):
                    member_rows=member_rows)
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
-        recv = self.comm.all_to_all_rows(payload)  # (world*r, shard)
+        recv = self.exchanged(payload)  # (r*world, shard)
         maxdiff = ops.pair_maxdiff(recv, self.pairs_a, self.pairs_b)  # (n_pairs,)
         if self.rtol > 0.0:
             rowmax = ops.row_absmax(recv)  # (world*r,)
@@ -156,7 +192,7 @@ class GeoMedianAggregator(Aggregator):
         self.tol = tol
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
-        recv = self.comm.all_to_all_rows(payload)  # (P, shard)
+        recv = self.exchanged(payload)  # (P, shard)
         P = recv.shape[0]
         z = recv.mean(dim=0)  # init at the mean (hdmedians does the same)
         z_new = torch.empty_like(z)
@@ -194,7 +230,7 @@ class KrumAggregator(Aggregator):
         self.s = s
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
-        recv = self.comm.all_to_all_rows(payload)  # (P, shard)
+        recv = self.exchanged(payload)  # (P, shard)
         P = recv.shape[0]
         gram = ops.segment_gram(recv, self.local_seg)  # (L, P, P)
         self.comm.all_reduce(gram)
@@ -241,20 +277,26 @@ class CyclicAggregator(Aggregator):
         self.code = code
         self.L = workers_per_rank
         self.n = code.n
+        world = max(comm.world, 1)
+        # worker w = l*world + src; its (re, im) payload rows at the source are
+        # (2l, 2l+1), so after the l-major all_to_all its recv rows are
+        # (2l+p)*world + src
+        w_ids = np.arange(self.n)
+        l, src = w_ids // world, w_ids % world
+        self.rows_re = (2 * l) * world + src
+        self.rows_im = (2 * l + 1) * world + src
 
     def aggregate(self, payload_planes: torch.Tensor, step: int) -> torch.Tensor:
-        # payload_planes: (L*2, d_pad); worker w = rank*L + l owns rows (2l, 2l+1)
-        recv = self.comm.all_to_all_rows(payload_planes)  # (world*L*2, shard)
-        r_planes = recv.view(self.n, 2, self.space.shard)
+        recv = self.exchanged(payload_planes)  # (2L*world, shard)
         gen = torch.Generator(device="cpu")
         gen.manual_seed(0x5EED ^ (step * 1000003) ^ self.comm.rank)
         z = torch.normal(
             mean=1.0, std=1.0, size=(self.space.shard,), generator=gen, dtype=torch.float32
         ).to(self.space.device)
-        proj = ops.cyclic_project(r_planes, z)  # (n, 2) partial
+        proj = ops.cyclic_project(recv, z)  # (2L*world,) per-row partial dots
         self.comm.all_reduce(proj)
-        proj_c = proj.to("cpu").numpy().astype(np.float64)
-        proj_complex = proj_c[:, 0] + 1j * proj_c[:, 1]
+        pa = proj.to("cpu").numpy().astype(np.float64)
+        proj_complex = pa[self.rows_re] + 1j * pa[self.rows_im]
         syndrome = self.code.W_perp @ proj_complex
         scale = float(np.abs(proj_complex).max())
         if float(np.abs(syndrome).max()) <= 1e-7 * max(scale, 1e-30):
@@ -262,9 +304,11 @@ class CyclicAggregator(Aggregator):
         else:
             healthy = self.code.locate_errors(syndrome)
         v = self.code.recombination_vector(healthy)
-        v_re = torch.tensor(np.real(v), dtype=torch.float32)
-        v_im = torch.tensor(np.imag(v), dtype=torch.float32)
-        ops.cyclic_recombine(r_planes, v_re.to(recv.device), v_im.to(recv.device), self._shard_out)
-        self._shard_out /= float(self.n)
+        # Re(v @ R) = sum_w vre[w]*re_row[w] - vim[w]*im_row[w]: one combine kernel
+        rows = torch.tensor(np.concatenate([self.rows_re, self.rows_im]),
+                            dtype=torch.int64, device=recv.device)
+        w = torch.tensor(np.concatenate([np.real(v), -np.imag(v)]) / self.n,
+                         dtype=torch.float32, device=recv.device)
+        ops.combine_rows(recv, rows, w, self._shard_out)
         self.comm.all_gather_shard(self._shard_out, self._out)
         return self._out

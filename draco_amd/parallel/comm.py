@@ -53,22 +53,35 @@ class Communicator:
             else:
                 dist.barrier()
 
-    def all_to_all_rows(self, payload: torch.Tensor) -> torch.Tensor:
-        """payload: (L, world * shard) -> received (world * L, shard).
+    def all_to_all_row(self, payload_row: torch.Tensor, recv_row: torch.Tensor,
+                       async_op: bool = False):
+        """One payload row (d_pad,) -> recv_row (world, shard): recv_row[src] is rank
+        src's shard-for-me of this row.  Zero-copy (each rank's chunk of a contiguous
+        flat row IS the send buffer); async_op returns the work handle so the exchange
+        overlaps the remaining backward compute.
+        """
+        if not self.distributed:
+            recv_row.view(-1).copy_(payload_row)
+            return None
+        work = dist.all_to_all_single(recv_row.view(-1), payload_row, async_op=async_op)
+        return work if async_op else None
 
-        Row w = src * L + l of the result is rank src's local worker l's shard for
-        this rank.  world == 1 degenerates to a reshape (no copy).
+    def all_to_all_rows(self, payload: torch.Tensor) -> torch.Tensor:
+        """payload: (L, world * shard) -> received (L * world, shard).
+
+        ROW CONVENTION: row l * world + src of the result is rank src's local row l's
+        shard for this rank (l-major).  world == 1 degenerates to a reshape (no copy).
         """
         L, d_pad = payload.shape
         shard = d_pad // self.world
         if not self.distributed:
             return payload.view(L, shard)
-        # (L, world, shard) -> (world, L, shard) so the send buffer is contiguous per
-        # destination rank
-        send = payload.view(L, self.world, shard).transpose(0, 1).contiguous()
-        recv = torch.empty_like(send)
-        dist.all_to_all_single(recv.view(-1), send.view(-1))
-        return recv.view(self.world * L, shard)
+        recv = torch.empty(L, self.world, shard, dtype=payload.dtype, device=payload.device)
+        works = [self.all_to_all_row(payload[l], recv[l], async_op=True) for l in range(L)]
+        for w in works:
+            if w is not None:
+                w.wait()
+        return recv.view(L * self.world, shard)
 
     def reduce_scatter_sum(self, payload_sum: torch.Tensor) -> torch.Tensor:
         """payload_sum: (d_pad,) local sum -> (shard,) global sum of this rank's shard."""

@@ -124,7 +124,7 @@ class Trainer:
             sup = self.code.support
             self._w_re, self._w_im = [], []
             for l in range(self.L):
-                w_global = self.rank * self.L + l
+                w_global = l * self.world + self.rank
                 coeff = W[w_global, sup[w_global]]
                 self._w_re.append(torch.tensor(np.real(coeff), dtype=torch.float32, device=device))
                 self._w_im.append(torch.tensor(np.imag(coeff), dtype=torch.float32, device=device))
@@ -181,14 +181,16 @@ class Trainer:
                     worker_id = self.rank
                 else:
                     group = (self.rank - l) % self.world
-                    worker_id = ((group + l) % self.world) * self.L + l  # == rank*L + l
+                    worker_id = l * self.world + self.rank  # l-major global worker id
                 x, y = self.data.batch_for(group, step)
                 losses.append(self._forward_backward(x, y, self.payload[l]))
                 if worker_id in adversaries:
                     ops.inject_(self.payload[l], cfg.err_mode, cyclic=False)
+                # overlap: this row's all_to_all runs while the next backward computes
+                self.agg.start_row(self.payload, l)
         else:  # cyclic
             for l in range(self.L):
-                w_global = self.rank * self.L + l
+                w_global = l * self.world + self.rank
                 sup = self.code.support[w_global]
                 for k in range(self.s_hat):
                     x, y = self.data.sub_batch(int(sup[k]), step)
@@ -197,6 +199,8 @@ class Trainer:
                 ops.cyclic_encode(self.scratch, self._w_re[l], self._w_im[l], enc)
                 if w_global in adversaries:
                     self._inject_encoded(enc, cfg.err_mode)
+                self.agg.start_row(self.payload, 2 * l)
+                self.agg.start_row(self.payload, 2 * l + 1)
 
         t_comp = time.perf_counter()
         grad = self.agg.aggregate(self.payload, step)

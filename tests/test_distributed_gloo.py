@@ -22,11 +22,11 @@ def _a2a_worker(rank, world):
     for l in range(L):
         payload[l] = rank * 100 + l * 10 + torch.arange(d_pad, dtype=torch.float32) / 100.0
     recv = comm.all_to_all_rows(payload)
-    # row w = src*L + l must hold src's worker-l shard for this rank
+    # l-major: row l*world + src must hold src's row-l shard for this rank
     for src in range(world):
         for l in range(L):
             expect = src * 100 + l * 10 + (torch.arange(shard) + rank * shard) / 100.0
-            assert torch.allclose(recv[src * L + l], expect.float()), (rank, src, l)
+            assert torch.allclose(recv[l * world + src], expect.float()), (rank, src, l)
     comm.shutdown()
     return True
 
@@ -171,7 +171,7 @@ def _cyclic_equiv_worker(rank, world):
 
     payload = space.alloc_payload(L * 2)
     for l in range(L):
-        w = rank * L + l
+        w = l * world + rank
         sup = code.support[w]
         grads = torch.stack([sub_grad(int(j)) for j in sup])
         wre = torch.tensor(np.real(code.W[w, sup]), dtype=torch.float32)

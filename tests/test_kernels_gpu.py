@@ -115,9 +115,18 @@ def test_cyclic_kernels():
 
     r = _r(n, 2, d, seed=13)
     z = _r(d, seed=14)
-    proj_ref = fb.cyclic_project(r, z)
-    proj_gpu = ops.cyclic_project(r.to(DEV).contiguous(), z.to(DEV))
+    r2 = r.view(2 * n, d)
+    proj_ref = fb.cyclic_project(r2, z)
+    proj_gpu = ops.cyclic_project(r2.to(DEV).contiguous(), z.to(DEV))
     assert torch.allclose(proj_ref, proj_gpu.cpu(), rtol=1e-4, atol=1e-2)
+
+    rows = torch.tensor([1, 4, 7, 10])
+    wr = _r(4, seed=21)
+    outc_ref = torch.zeros(d)
+    fb.combine_rows(r2, rows, wr, outc_ref)
+    outc_gpu = torch.zeros(d, device=DEV)
+    ops.combine_rows(r2.to(DEV).contiguous(), rows.to(DEV), wr.to(DEV), outc_gpu)
+    assert torch.allclose(outc_ref, outc_gpu.cpu(), atol=1e-3)
 
     vre, vim = _r(n, seed=15), _r(n, seed=16)
     out_ref = torch.zeros(d)

@@ -95,10 +95,17 @@ def test_cyclic_ops_match_numpy():
     ref = (g.numpy() * wre.numpy()[:, None]).sum(0)
     assert np.allclose(out[0].numpy(), ref, atol=1e-5)
 
-    r = torch.tensor(rng.normal(size=(n, 2, d)), dtype=torch.float32)
+    r = torch.tensor(rng.normal(size=(2 * n, d)), dtype=torch.float32)
     z = torch.tensor(rng.normal(size=d), dtype=torch.float32)
     proj = fb.cyclic_project(r, z)
     assert np.allclose(proj.numpy(), (r.numpy() @ z.numpy()), atol=1e-4)
+
+    rows = torch.tensor([0, 3, 5])
+    w = torch.tensor(rng.normal(size=3), dtype=torch.float32)
+    outc = torch.zeros(d)
+    fb.combine_rows(r, rows, w, outc)
+    assert np.allclose(outc.numpy(), w.numpy() @ r.numpy()[rows.numpy()], atol=1e-4)
+    r = r.reshape(n, 2, d)
 
     vre = torch.tensor(rng.normal(size=n), dtype=torch.float32)
     vim = torch.tensor(rng.normal(size=n), dtype=torch.float32)
