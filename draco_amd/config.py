@@ -43,6 +43,7 @@ class Config:
     dtype: str = "bf16"               # bf16|fp32 compute dtype (grads/aggregation always fp32)
     device: str = "auto"              # auto|cuda|cpu
     deterministic: bool = False
+    channels_last: bool = True    # NHWC conv layout on GPU (no effect on CPU)
     log_dir: str = "output/logs/"
 
     def sanity(self):

@@ -31,9 +31,16 @@ def _pad(d: int, world: int) -> int:
 
 
 class FlatSpace:
-    def __init__(self, model: torch.nn.Module, world: int, device: torch.device):
+    """channels_last=True stores 4D (conv-weight) parameters NHWC inside the flat
+    buffer and re-points p.data at a permuted view with channels_last strides, so
+    MIOpen picks its NHWC kernels (+7% measured on ResNet-18 fwd+bwd) while the flat
+    comm/decode space stays one contiguous fp32 range."""
+
+    def __init__(self, model: torch.nn.Module, world: int, device: torch.device,
+                 channels_last: bool = False):
         self.device = device
         self.world = world
+        self.channels_last = channels_last
         params = [p for p in model.parameters() if p.requires_grad]
         self.params = params
         self.shapes = [p.shape for p in params]
@@ -50,11 +57,24 @@ class FlatSpace:
         self.flat_param = torch.zeros(self.d_pad, dtype=torch.float32, device=device)
         with torch.no_grad():
             for p, o, n in zip(params, self.offsets, self.numels):
-                self.flat_param[o : o + n].copy_(p.data.reshape(-1).to(device=device, dtype=torch.float32))
-                p.data = self.flat_param[o : o + n].view(p.shape)
+                if self._is_cl(p):
+                    src = p.data.to(device=device, dtype=torch.float32)
+                    self.flat_param[o : o + n].copy_(src.permute(0, 2, 3, 1).reshape(-1))
+                else:
+                    self.flat_param[o : o + n].copy_(p.data.reshape(-1).to(device=device, dtype=torch.float32))
+                p.data = self._view(self.flat_param, o, n, p.shape)
 
         # (L+1) segment bounds in global flat coordinates (one segment per parameter)
         self.seg_bounds = torch.tensor(self.offsets + [self.d], dtype=torch.int64)
+
+    def _is_cl(self, p) -> bool:
+        return self.channels_last and p.dim() == 4
+
+    def _view(self, buf, o, n, shape):
+        if self.channels_last and len(shape) == 4:
+            O, C, H, W = shape
+            return buf[o : o + n].view(O, H, W, C).permute(0, 3, 1, 2)
+        return buf[o : o + n].view(shape)
 
     # ------------------------------------------------------------------ grads
     def alloc_payload(self, rows: int) -> torch.Tensor:
@@ -63,8 +83,8 @@ class FlatSpace:
     def attach_grads(self, buf: torch.Tensor) -> None:
         """Point every parameter's .grad at views of the given flat (d_pad,) buffer."""
         assert buf.shape == (self.d_pad,)
-        for p, o, n in zip(self.params, self.offsets, self.numels):
-            p.grad = buf[o : o + n].view(p.shape)
+        for p, o, n, shape in zip(self.params, self.offsets, self.numels, self.shapes):
+            p.grad = self._view(buf, o, n, shape)
 
     def detach_grads(self) -> None:
         for p in self.params:

@@ -73,7 +73,8 @@ class Trainer:
         torch.manual_seed(cfg.seed)
         self.model = build_model(cfg.network, cfg.dataset).to(device)
         self.model.train()
-        self.space = FlatSpace(self.model, self.world, device)
+        self.use_cl = cfg.channels_last and device.type == "cuda"
+        self.space = FlatSpace(self.model, self.world, device, channels_last=self.use_cl)
         self.opt = FlatSGD(self.space.flat_param, lr=cfg.lr, momentum=cfg.momentum)
 
         self.autocast_dtype = torch.bfloat16 if (cfg.dtype == "bf16" and device.type == "cuda") else None
@@ -154,6 +155,8 @@ class Trainer:
         raise ValueError(f"baseline approach supports modes normal/geometric_median/krum, got {cfg.mode!r}")
 
     def _forward_backward(self, x, y, grad_row: torch.Tensor) -> float:
+        if self.use_cl and x.dim() == 4:
+            x = x.to(memory_format=torch.channels_last)
         self.space.attach_grads(grad_row)
         grad_row.zero_()
         if self.autocast_dtype is not None:
