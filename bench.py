@@ -37,7 +37,8 @@ def main():
     p.add_argument("--err-mode", type=str, default="rev_grad")
     p.add_argument("--dtype", type=str, default="bf16")
     p.add_argument("--device", type=str, default="auto")
-    p.add_argument("--channels-last", type=lambda v: v.lower() in ("1","true"), default=True)
+    p.add_argument("--channels-last", type=lambda v: v.lower() in ("1","true"), default=False)
+    p.add_argument("--hip-graphs", type=lambda v: v.lower() in ("1","true"), default=True)
     args = p.parse_args()
 
     from draco_amd.config import Config
@@ -64,6 +65,7 @@ def main():
         train_dir="gpurun_out/bench_ckpt",
         topology="colocated",
         channels_last=args.channels_last,
+        hip_graphs=args.hip_graphs,
     )
     t = Trainer(cfg)
     t.logger.stdout_every = 0

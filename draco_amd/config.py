@@ -43,7 +43,8 @@ class Config:
     dtype: str = "bf16"               # bf16|fp32 compute dtype (grads/aggregation always fp32)
     device: str = "auto"              # auto|cuda|cpu
     deterministic: bool = False
-    channels_last: bool = True    # NHWC conv layout on GPU (no effect on CPU)
+    channels_last: bool = False   # NHWC conv layout (measured net-negative with MIOpen wrw on flat views)
+    hip_graphs: bool = True       # capture fwd+bwd in hipGraphs (launch-bound models; GPU only)
     log_dir: str = "output/logs/"
 
     def sanity(self):
