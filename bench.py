@@ -99,7 +99,9 @@ def main():
 
     if rank == 0:
         result = {
-            "metric": "images/sec (effective), ResNet-18 CIFAR-10, repetition r=3, s=1 adversary",
+            "metric": (f"images/sec (effective), {args.network} {args.dataset}, "
+                       f"{args.approach} r={args.group_size if args.approach == 'maj_vote' else 2 * args.worker_fail + 1}, "
+                       f"s={args.worker_fail} adversary"),
             "value": round(images_per_sec, 2),
             "unit": "images/sec",
             "n_gpus": world,

@@ -22,10 +22,25 @@ def get_extension():
         _tried = True
         try:
             _ext = importlib.import_module("draco_amd._hip_ops")
+            _check_stale(_ext)
         except Exception as e:  # pragma: no cover - exercised on GPU boxes
             _ext = None
             _err = e
     return _ext
+
+
+def _check_stale(ext) -> None:
+    import warnings
+
+    src = os.path.join(os.path.dirname(__file__), "csrc", "draco_kernels.hip")
+    try:
+        if os.path.getmtime(src) > os.path.getmtime(ext.__file__) + 1.0:
+            warnings.warn(
+                "draco_amd._hip_ops is OLDER than its kernel source — rebuild with "
+                "`python setup.py build_ext --inplace` (stale .so gives wrong/missing ops)"
+            )
+    except OSError:
+        pass
 
 
 def require_extension():

@@ -321,7 +321,26 @@ class Trainer:
         step = self.step_num if step is None else step
         return os.path.join(self.cfg.train_dir, f"model_step_{step}")
 
+    def sync_buffers(self):
+        """Average BN running stats across ranks (they drift: each rank normalises its
+        own group batches; parameters stay bit-identical by construction)."""
+        if not self.comm.distributed:
+            return
+        bufs = [b for b in self.model.buffers() if b.dtype.is_floating_point]
+        if not bufs:
+            return
+        flat = torch.cat([b.reshape(-1).float() for b in bufs])
+        self.comm.all_reduce(flat)
+        flat /= self.world
+        off = 0
+        with torch.no_grad():
+            for b in bufs:
+                n = b.numel()
+                b.copy_(flat[off : off + n].view(b.shape).to(b.dtype))
+                off += n
+
     def save(self):
+        self.sync_buffers()
         save_checkpoint(self._ckpt_path(), self.model, self.space, self.opt, self.step_num, self.cfg)
 
     def load(self, step: int):
