@@ -55,7 +55,7 @@ class CyclicCode:
         return 2 * self.s + 1
 
     # ---------------------------------------------------------------- decode
-    def locate_errors(self, syndrome_proj: np.ndarray) -> np.ndarray:
+    def locate_errors(self, syndrome_proj: np.ndarray, known_bad=()) -> np.ndarray:
         """Return sorted healthy worker indices from a projected syndrome.
 
         syndrome_proj: (2s,) complex — W_perp @ (R @ z) for random projection z.
@@ -92,10 +92,13 @@ class CyclicCode:
         poly[s] = 1.0
         est = np.abs(self.Z @ poly)  # |a(z_i)|; ~0 at adversarial i
         bad = np.nonzero(est <= 1e-5 * max(est.max(), 1e-300))[0]
+        known = np.asarray(sorted(known_bad), dtype=np.int64)
+        if len(known):  # erasures (e.g. straggler timeouts) are errors at KNOWN rows
+            bad = np.union1d(bad, known)
         if 0 < len(bad) <= s and self._verify_bad(bad, syn, syn_norm):
             pass
         else:
-            bad = self._search_bad(syn, syn_norm)
+            bad = self._search_bad(syn, syn_norm, known)
         mask = np.ones(n, dtype=bool)
         mask[bad] = False
         return np.nonzero(mask)[0]
@@ -106,14 +109,19 @@ class CyclicCode:
         resid = np.linalg.norm(B @ eps - syn)
         return resid <= 1e-5 * max(syn_norm, 1e-300)
 
-    def _search_bad(self, syn: np.ndarray, syn_norm: float) -> np.ndarray:
-        """Exhaustive sparse-recovery fallback: smallest-residual support of size <= s."""
+    def _search_bad(self, syn: np.ndarray, syn_norm: float, known=()) -> np.ndarray:
+        """Exhaustive sparse-recovery fallback: smallest-residual support of size <= s
+        (always containing the known erasure locations)."""
         import itertools
 
-        best, best_resid = np.empty(0, dtype=np.int64), syn_norm
-        for k in range(1, self.s + 1):
-            for subset in itertools.combinations(range(self.n), k):
-                idx = np.asarray(subset, dtype=np.int64)
+        known = np.asarray(sorted(known), dtype=np.int64)
+        rest = [i for i in range(self.n) if i not in set(known.tolist())]
+        best, best_resid = known, syn_norm
+        for k in range(0, self.s - len(known) + 1):
+            if k == 0 and len(known) == 0:
+                continue
+            for subset in itertools.combinations(rest, k):
+                idx = np.union1d(known, np.asarray(subset, dtype=np.int64)).astype(np.int64)
                 B = self.W_perp[:, idx]
                 eps, *_ = np.linalg.lstsq(B, syn, rcond=None)
                 resid = np.linalg.norm(B @ eps - syn)

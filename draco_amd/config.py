@@ -45,6 +45,8 @@ class Config:
     deterministic: bool = False
     channels_last: bool = False   # NHWC conv layout (measured net-negative with MIOpen wrw on flat views)
     hip_graphs: bool = True       # capture fwd+bwd in hipGraphs (launch-bound models; GPU only)
+    straggler_timeout: float = 0.0  # ps topology: seconds after first gradient before
+                                    # missing workers become erasures (0 = wait forever)
     log_dir: str = "output/logs/"
 
     def sanity(self):

@@ -286,7 +286,8 @@ class CyclicAggregator(Aggregator):
         self.rows_re = (2 * l) * world + src
         self.rows_im = (2 * l + 1) * world + src
 
-    def aggregate(self, payload_planes: torch.Tensor, step: int) -> torch.Tensor:
+    def aggregate(self, payload_planes: torch.Tensor, step: int,
+                  erasures: frozenset = frozenset()) -> torch.Tensor:
         recv = self.exchanged(payload_planes)  # (2L*world, shard)
         gen = torch.Generator(device="cpu")
         gen.manual_seed(0x5EED ^ (step * 1000003) ^ self.comm.rank)
@@ -299,10 +300,10 @@ class CyclicAggregator(Aggregator):
         proj_complex = pa[self.rows_re] + 1j * pa[self.rows_im]
         syndrome = self.code.W_perp @ proj_complex
         scale = float(np.abs(proj_complex).max())
-        if float(np.abs(syndrome).max()) <= 1e-7 * max(scale, 1e-30):
+        if not erasures and float(np.abs(syndrome).max()) <= 1e-7 * max(scale, 1e-30):
             healthy = np.arange(self.n)
         else:
-            healthy = self.code.locate_errors(syndrome)
+            healthy = self.code.locate_errors(syndrome, known_bad=erasures)
         v = self.code.recombination_vector(healthy)
         # Re(v @ R) = sum_w vre[w]*re_row[w] - vim[w]*im_row[w]: one combine kernel
         rows = torch.tensor(np.concatenate([self.rows_re, self.rows_im]),

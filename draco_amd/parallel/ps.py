@@ -136,17 +136,21 @@ class Master(_PSBase):
     def run(self, max_steps: int | None = None):
         cfg = self.cfg
         steps = max_steps or cfg.max_steps
+        ctrl = torch.zeros(2, dtype=torch.int64)
         for _ in range(steps):
             t0 = time.perf_counter()
+            # step announce (reference tag-10 broadcast, baseline_master.py:156-162) +
+            # abort flag (the reference's vestigial tag-77 kill channel, done properly)
+            ctrl[0] = self.step_num
+            ctrl[1] = 0
+            self.comm.broadcast(ctrl, src=0)
             self.comm.broadcast(self.space.flat_param, src=0)
-            reqs = [
-                self.comm.irecv(self.gather_buf[w * self.payload_rows : (w + 1) * self.payload_rows], src=w + 1)
-                for w in range(self.P)
-            ]
-            for r in reqs:
-                r.wait()
+            erasures = self._gather_grads()
             t_gather = time.perf_counter()
-            grad = self.agg.aggregate(self.gather_buf, self.step_num)
+            if erasures and self.approach == "cyclic":
+                grad = self.agg.aggregate(self.gather_buf, self.step_num, erasures=erasures)
+            else:
+                grad = self.agg.aggregate(self.gather_buf, self.step_num)
             t_agg = time.perf_counter()
             self.opt.step(grad)
             t1 = time.perf_counter()
@@ -161,7 +165,82 @@ class Master(_PSBase):
                     os.path.join(cfg.train_dir, f"model_step_{self.step_num}"),
                     self.model, self.space, self.opt, self.step_num, cfg,
                 )
+        # tell workers to stop (clean shutdown; replaces the reference's missing
+        # master-side kill sender, SURVEY §2.3 straggler row)
+        ctrl[0] = self.step_num
+        ctrl[1] = 1
+        self.comm.broadcast(ctrl, src=0)
         self.logger.close()
+
+    def _gather_grads(self) -> frozenset:
+        """Gather one gradient payload per worker; with straggler_timeout > 0, give up
+        on workers that have not arrived within the timeout after the first arrival
+        and treat them as ERASURES (the cyclic decode tolerates them as known error
+        locations; a missing vote member simply loses its group's vote).
+
+        Protocol care: a timed-out worker's send still completes later (matched by the
+        orphaned irecv).  The request is kept in self._stale[w]; at the next step the
+        stale payload is DISCARDED and a fresh irecv posted, so the per-step message
+        stream stays aligned.  A worker whose stale request never completes is treated
+        as dead and stays erased (its irecv is never reposted)."""
+        timeout = self.cfg.straggler_timeout
+        if timeout <= 0:
+            reqs = [
+                self.comm.irecv(self.gather_buf[w * self.payload_rows : (w + 1) * self.payload_rows], src=w + 1)
+                for w in range(self.P)
+            ]
+            for r in reqs:
+                r.wait()
+            return frozenset()
+        # gloo Work.is_completed() never fires for p2p recvs without wait(), so
+        # completion is observed through one waiter thread per outstanding request
+        # (blocking wait() + Event; threads are daemonic and bounded by P).
+        import threading
+
+        if not hasattr(self, "_stale"):
+            self._stale = {}
+        events = {}
+        dead = set()
+        for w in range(self.P):
+            buf = self.gather_buf[w * self.payload_rows : (w + 1) * self.payload_rows]
+            old = self._stale.get(w)
+            if old is not None:
+                if old.is_set():
+                    del self._stale[w]  # late (previous-step) payload arrived: discard
+                else:
+                    dead.add(w)  # still missing: remains erased, no new irecv
+                    continue
+            req = self.comm.irecv(buf, src=w + 1)
+            ev = threading.Event()
+
+            def waiter(r=req, e=ev):
+                try:
+                    r.wait()
+                finally:
+                    e.set()
+
+            threading.Thread(target=waiter, daemon=True).start()
+            events[w] = ev
+        deadline = None
+        pending = set(events)
+        while pending:
+            done = {w for w in pending if events[w].is_set()}
+            pending -= done
+            if not pending:
+                break
+            now = time.perf_counter()
+            if deadline is None:
+                if done or dead:
+                    deadline = now + timeout
+            elif now > deadline:
+                break
+            time.sleep(0.0005)
+        for w in pending:
+            self._stale[w] = events[w]
+        missing = frozenset(pending | dead)
+        for w in missing:
+            self.gather_buf[w * self.payload_rows : (w + 1) * self.payload_rows].zero_()
+        return missing
 
 
 class Worker(_PSBase):
@@ -188,8 +267,13 @@ class Worker(_PSBase):
     def run(self, max_steps: int | None = None):
         cfg = self.cfg
         steps = max_steps or cfg.max_steps
-        for _ in range(steps):
+        ctrl = torch.zeros(2, dtype=torch.int64)
+        for _ in range(steps + 1):
             t0 = time.perf_counter()
+            self.comm.broadcast(ctrl, src=0)
+            if int(ctrl[1]) != 0:
+                break  # master sent the kill signal
+            self.step_num = int(ctrl[0])
             self.comm.broadcast(self.space.flat_param, src=0)
             adversaries = self.schedule.adversaries_at(self.step_num)
             if cfg.approach == "cyclic":
@@ -214,7 +298,6 @@ class Worker(_PSBase):
                 "step": self.step_num, "role": "worker", "loss": loss,
                 "comp": t_comp - t0, "comm": t1 - t_comp, "time": t1 - t0,
             })
-            self.step_num += 1
         self.logger.close()
 
     def _fwd_bwd(self, x, y, row):
