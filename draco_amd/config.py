@@ -30,8 +30,10 @@ class Config:
     adversarial: int = 1              # parity flag (err magnitude switch; reference hardcodes -100)
     worker_fail: int = 2
     group_size: int = 5
-    compress_grad: str = "none"       # none|bf16 (blosc replaced by GPU dtype cast)
+    compress_grad: str = "none"       # none|bf16 wire dtype for gradient exchange (replaces the
+                                      # reference's blosc/snappy CPU compression)
     checkpoint_step: int = 0
+    optimizer: str = "sgd"            # sgd|adam (fused flat kernels, optim/flat_sgd.py)
 
     # MI355X-native extensions
     topology: str = "colocated"       # colocated (all ranks compute) | ps (rank0 = parameter server)

@@ -34,13 +34,16 @@ class Aggregator:
 
     name = "base"
 
-    def __init__(self, comm: Communicator, space: FlatSpace):
+    def __init__(self, comm: Communicator, space: FlatSpace, comm_dtype=torch.float32):
         self.comm = comm
         self.space = space
+        self.comm_dtype = comm_dtype  # bf16 = K12 wire compression (all_to_all paths)
         self._out = torch.zeros(space.d_pad, dtype=torch.float32, device=space.device)
         self._shard_out = torch.zeros(space.shard, dtype=torch.float32, device=space.device)
         self.local_seg = space.local_seg_bounds(comm.rank).to(space.device)
         self._recv = None
+        self._send = None
+        self._recv32 = None
         self._works: list = []
         self._started: set = set()
 
@@ -50,8 +53,18 @@ class Aggregator:
             return
         if self._recv is None or self._recv.shape[0] != payload.shape[0]:
             self._recv = torch.empty(payload.shape[0], self.comm.world, self.space.shard,
-                                     dtype=payload.dtype, device=payload.device)
-        w = self.comm.all_to_all_row(payload[row], self._recv[row], async_op=True)
+                                     dtype=self.comm_dtype, device=payload.device)
+            if self.comm_dtype != torch.float32:
+                self._send = torch.empty(payload.shape[0], payload.shape[1],
+                                         dtype=self.comm_dtype, device=payload.device)
+                self._recv32 = torch.empty(payload.shape[0], self.comm.world, self.space.shard,
+                                           dtype=torch.float32, device=payload.device)
+        if self.comm_dtype != torch.float32:
+            self._send[row].copy_(payload[row])  # fp32 -> bf16 cast on device
+            send_row = self._send[row]
+        else:
+            send_row = payload[row]
+        w = self.comm.all_to_all_row(send_row, self._recv[row], async_op=True)
         if w is not None:
             self._works.append(w)
         self._started.add(row)
@@ -68,6 +81,9 @@ class Aggregator:
             w.wait()
         self._works = []
         self._started = set()
+        if self.comm_dtype != torch.float32:
+            self._recv32.copy_(self._recv)  # upcast once; decode kernels stay fp32
+            return self._recv32.view(rows * self.comm.world, self.space.shard)
         return self._recv.view(rows * self.comm.world, self.space.shard)
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
@@ -116,8 +132,8 @@ class VoteAggregator(Aggregator):
     name = "maj_vote"
 
     def __init__(self, comm, space, group_size: int, atol: float = 0.0,
-                 rtol: float = 0.0, member_rows=None):
-        super().__init__(comm, space)
+                 rtol: float = 0.0, member_rows=None, comm_dtype=torch.float32):
+        super().__init__(comm, space, comm_dtype)
         self.atol = atol
         self.rtol = rtol
         if member_rows is None:
@@ -272,8 +288,9 @@ class CyclicAggregator(Aggregator):
 
     name = "cyclic"
 
-    def __init__(self, comm, space, code: CyclicCode, workers_per_rank: int):
-        super().__init__(comm, space)
+    def __init__(self, comm, space, code: CyclicCode, workers_per_rank: int,
+                 comm_dtype=torch.float32):
+        super().__init__(comm, space, comm_dtype)
         self.code = code
         self.L = workers_per_rank
         self.n = code.n

@@ -136,7 +136,8 @@ class Master(_PSBase):
     def run(self, max_steps: int | None = None):
         cfg = self.cfg
         steps = max_steps or cfg.max_steps
-        ctrl = torch.zeros(2, dtype=torch.int64)
+        ctrl = torch.zeros(2, dtype=torch.int64,
+                           device=self.device if self.comm.backend == "nccl" else "cpu")
         for _ in range(steps):
             t0 = time.perf_counter()
             # step announce (reference tag-10 broadcast, baseline_master.py:156-162) +
@@ -267,7 +268,8 @@ class Worker(_PSBase):
     def run(self, max_steps: int | None = None):
         cfg = self.cfg
         steps = max_steps or cfg.max_steps
-        ctrl = torch.zeros(2, dtype=torch.int64)
+        ctrl = torch.zeros(2, dtype=torch.int64,
+                           device=self.device if self.comm.backend == "nccl" else "cpu")
         for _ in range(steps + 1):
             t0 = time.perf_counter()
             self.comm.broadcast(ctrl, src=0)

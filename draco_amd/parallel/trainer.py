@@ -75,7 +75,12 @@ class Trainer:
         self.model.train()
         self.use_cl = cfg.channels_last and device.type == "cuda"
         self.space = FlatSpace(self.model, self.world, device, channels_last=self.use_cl)
-        self.opt = FlatSGD(self.space.flat_param, lr=cfg.lr, momentum=cfg.momentum)
+        if cfg.optimizer == "adam":
+            from ..optim import FlatAdam
+
+            self.opt = FlatAdam(self.space.flat_param, lr=cfg.lr)
+        else:
+            self.opt = FlatSGD(self.space.flat_param, lr=cfg.lr, momentum=cfg.momentum)
 
         self.autocast_dtype = torch.bfloat16 if (cfg.dtype == "bf16" and device.type == "cuda") else None
 
@@ -106,7 +111,8 @@ class Trainer:
                     rtol = 1e-4  # fp32 reorder noise
             self.vote_rtol = rtol
             self.agg = VoteAggregator(self.comm, self.space, group_size=self.r,
-                                      atol=cfg.vote_atol, rtol=rtol)
+                                      atol=cfg.vote_atol, rtol=rtol,
+                                      comm_dtype=self._comm_dtype())
             self.data = GroupBatchSource(self._dataset(), cfg.batch_size, n_groups=self.G)
             self.payload = self.space.alloc_payload(self.L)
         elif approach == "cyclic":
@@ -116,7 +122,8 @@ class Trainer:
             self.P = self.n
             self.code = build_cyclic_code(self.n, cfg.worker_fail)
             self.s_hat = self.code.s_hat
-            self.agg = CyclicAggregator(self.comm, self.space, self.code, self.L)
+            self.agg = CyclicAggregator(self.comm, self.space, self.code, self.L,
+                                        comm_dtype=self._comm_dtype())
             self.data = GlobalBatchSource(self._dataset(), cfg.batch_size, n_workers=self.n)
             # encoded complex payload planes + raw sub-batch gradient scratch
             self.payload = self.space.alloc_payload(self.L * 2)
@@ -145,6 +152,9 @@ class Trainer:
             self.load(cfg.checkpoint_step)
 
     # ------------------------------------------------------------------ helpers
+    def _comm_dtype(self):
+        return torch.bfloat16 if self.cfg.compress_grad == "bf16" else torch.float32
+
     def _dataset(self) -> SyntheticClassification:
         return SyntheticClassification(self.cfg.dataset, self.device, seed=1234)
 

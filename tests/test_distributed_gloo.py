@@ -189,3 +189,30 @@ def _cyclic_equiv_worker(rank, world):
 
 def test_sharded_cyclic_equals_local():
     run_dist(_cyclic_equiv_worker, 2)
+
+
+def _compress_worker(rank, world):
+    """bf16 wire compression: vote result must match fp32 within bf16 tolerance."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    out = {}
+    for comp in ("none", "bf16"):
+        cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                     approach="maj_vote", mode="maj_vote", group_size=3, worker_fail=1,
+                     compress_grad=comp, max_steps=50, eval_freq=0, log_dir="",
+                     train_dir="/tmp/draco_comp")
+        t = Trainer(cfg)
+        t.logger.stdout_every = 0
+        for _ in range(4):
+            t.train_step()
+        out[comp] = t.space.flat_param.clone()
+        t.close()
+    diff = (out["none"] - out["bf16"]).abs().max()
+    scale = out["none"].abs().max()
+    assert diff < 0.02 * scale, float(diff)
+    return True
+
+
+def test_bf16_wire_compression():
+    run_dist(_compress_worker, 2)
