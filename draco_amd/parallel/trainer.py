@@ -141,7 +141,7 @@ class Trainer:
 
         self.use_graphs = bool(cfg.hip_graphs) and device.type == "cuda" and not cfg.deterministic
         self._graphs = {}
-        self._graph_pool = None
+        self._streams = []
         self.n_fail = min(cfg.worker_fail, self.P)
         self.schedule = AdversarySchedule(self.P, self.n_fail, cfg.max_steps)
         self.step_num = 0
@@ -188,13 +188,20 @@ class Trainer:
     # the whole thing as one launch.  The flat spaces make this natural: the payload
     # row IS a static buffer, so the captured backward accumulates straight into the
     # comm buffer; batch data is copied into static inputs before each replay.
-    def _run_fwd_bwd(self, key, grad_row: torch.Tensor, x, y):
+    # Concurrent replay: the L logical workers' graphs are mutually independent
+    # (private mempools, disjoint payload rows, shared READ-ONLY params), and a
+    # 128-image CIFAR conv underfills 256 CUs — so each worker's graph replays on its
+    # own HIP stream and the three fwd+bwd overlap on the chip.  BN running-stat
+    # writes would race across replicas, so only graph 0 updates them (the other
+    # graphs are captured with BN momentum frozen to 0, making their stat writes
+    # read-modify-write identities; stats are eval-only and rank-averaged anyway).
+    def _run_fwd_bwd(self, key, grad_row: torch.Tensor, x, y, stream=None):
         if not self.use_graphs:
             return self._forward_backward(x, y, grad_row)
         g = self._graphs.get(key)
         if g is None:
             try:
-                g = self._capture(grad_row, x, y)
+                g = self._capture(grad_row, x, y, freeze_bn_stats=key[1] != 0)
             except RuntimeError as e:  # pragma: no cover - capture unsupported
                 import warnings
 
@@ -202,12 +209,26 @@ class Trainer:
                 self.use_graphs = False
                 return self._forward_backward(x, y, grad_row)
             self._graphs[key] = g
-        g["x"].copy_(x)
-        g["y"].copy_(y)
-        g["graph"].replay()
+        if stream is not None:
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                g["x"].copy_(x)
+                g["y"].copy_(y)
+                g["graph"].replay()
+        else:
+            g["x"].copy_(x)
+            g["y"].copy_(y)
+            g["graph"].replay()
         return g["loss"]
 
-    def _capture(self, grad_row: torch.Tensor, x, y):
+    def _worker_stream(self, idx: int):
+        if not self.use_graphs:
+            return None
+        while len(self._streams) <= idx:
+            self._streams.append(torch.cuda.Stream())
+        return self._streams[idx]
+
+    def _capture(self, grad_row: torch.Tensor, x, y, freeze_bn_stats: bool = False):
         self.space.attach_grads(grad_row)
         static_x = x.clone()
         static_y = y.clone()
@@ -222,19 +243,29 @@ class Trainer:
             loss.backward()
             return loss
 
-        # warm up on a side stream (MIOpen find, autograd graph materialisation)
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):
-            for _ in range(2):
-                body()
-        torch.cuda.current_stream().wait_stream(s)
+        bn_mom = []
+        if freeze_bn_stats:
+            for m in self.model.modules():
+                if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+                    bn_mom.append((m, m.momentum))
+                    m.momentum = 0.0  # frozen into the captured BN kernels
+        try:
+            # warm up on a side stream (MIOpen find, autograd graph materialisation)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    body()
+            torch.cuda.current_stream().wait_stream(s)
 
-        if self._graph_pool is None:
-            self._graph_pool = torch.cuda.graph_pool_handle()
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph, pool=self._graph_pool):
-            loss = body()
+            # private mempool per graph: graphs sharing a pool may alias activation
+            # memory, which forbids concurrent replay
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                loss = body()
+        finally:
+            for m, mom in bn_mom:
+                m.momentum = mom
         return {"graph": graph, "x": static_x, "y": static_y, "loss": loss.detach()}
 
     # ------------------------------------------------------------------ one step
@@ -246,6 +277,7 @@ class Trainer:
         adversaries = self.schedule.adversaries_at(step) if self.n_fail > 0 else frozenset()
 
         if self.approach in ("baseline", "maj_vote"):
+            pending = []
             for l in range(self.L):
                 if self.approach == "baseline":
                     group = self.rank  # every worker draws its own stream
@@ -254,18 +286,29 @@ class Trainer:
                     group = (self.rank - l) % self.world
                     worker_id = l * self.world + self.rank  # l-major global worker id
                 x, y = self.data.batch_for(group, step)
-                losses.append(self._run_fwd_bwd(("slot", l), self.payload[l], x, y))
+                st = self._worker_stream(l)
+                losses.append(self._run_fwd_bwd(("slot", l), self.payload[l], x, y, stream=st))
+                pending.append((l, worker_id, st))
+            for l, worker_id, st in pending:
+                if st is not None:  # order the default stream after worker l's graph
+                    torch.cuda.current_stream().wait_stream(st)
                 if worker_id in adversaries:
                     ops.inject_(self.payload[l], cfg.err_mode, cyclic=False)
-                # overlap: this row's all_to_all runs while the next backward computes
+                # overlap: this row's all_to_all runs while other backwards compute
                 self.agg.start_row(self.payload, l)
         else:  # cyclic
             for l in range(self.L):
                 w_global = l * self.world + self.rank
                 sup = self.code.support[w_global]
+                streams = []
                 for k in range(self.s_hat):
                     x, y = self.data.sub_batch(int(sup[k]), step)
-                    losses.append(self._run_fwd_bwd(("sub", k), self.scratch[k], x, y))
+                    st = self._worker_stream(k)
+                    losses.append(self._run_fwd_bwd(("sub", k), self.scratch[k], x, y, stream=st))
+                    streams.append(st)
+                for st in streams:
+                    if st is not None:
+                        torch.cuda.current_stream().wait_stream(st)
                 enc = self.payload[2 * l : 2 * l + 2]
                 ops.cyclic_encode(self.scratch, self._w_re[l], self._w_im[l], enc)
                 if w_global in adversaries:
