@@ -114,3 +114,20 @@ def test_evaluate_runs(tmp_path):
     m = t.evaluate(n_batches=2)
     assert 0.0 <= m["prec1"] <= 1.0
     t.close()
+
+
+def test_evaluator_checkpoint(tmp_path):
+    """Standalone evaluator consumes the model_step_N layout (distributed_evaluator
+    parity)."""
+    from draco_amd.evaluate import evaluate_checkpoint
+
+    cfg = _cfg(tmp_path, approach="baseline", mode="normal", worker_fail=0, eval_freq=4)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    for _ in range(4):
+        t.train_step()
+    t.close()
+    path = os.path.join(cfg.train_dir, "model_step_4")
+    assert os.path.exists(path)
+    rec = evaluate_checkpoint(path, torch.device("cpu"), batches=2)
+    assert rec["step"] == 4 and 0.0 <= rec["prec1"] <= 1.0 and 0.0 <= rec["prec5"] <= 1.0
