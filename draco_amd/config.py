@@ -49,6 +49,8 @@ class Config:
     hip_graphs: bool = True       # capture fwd+bwd in hipGraphs (launch-bound models; GPU only)
     straggler_timeout: float = 0.0  # ps topology: seconds after first gradient before
                                     # missing workers become erasures (0 = wait forever)
+    nan_guard: bool = True        # failure detection: skip updates on non-finite decode
+    gpu_timing: bool = False      # device-accurate phase spans via HIP events (metrics)
     log_dir: str = "output/logs/"
 
     def sanity(self):

@@ -152,8 +152,16 @@ class Trainer:
             self.load(cfg.checkpoint_step)
 
     # ------------------------------------------------------------------ helpers
+    def _event(self):
+        if self.device.type != "cuda" or not self.cfg.gpu_timing:
+            return None
+        ev = torch.cuda.Event(enable_timing=True)
+        ev.record()
+        return ev
+
     def _comm_dtype(self):
-        return torch.bfloat16 if self.cfg.compress_grad == "bf16" else torch.float32
+        # "compress" = reference CLI alias (README.md:118, blosc) -> bf16 wire dtype
+        return torch.bfloat16 if self.cfg.compress_grad in ("bf16", "compress") else torch.float32
 
     def _dataset(self) -> SyntheticClassification:
         return SyntheticClassification(self.cfg.dataset, self.device, seed=1234)
@@ -273,6 +281,7 @@ class Trainer:
         cfg = self.cfg
         step = self.step_num
         t0 = time.perf_counter()
+        self._ev_step = self._event()
         losses = []
         adversaries = self.schedule.adversaries_at(step) if self.n_fail > 0 else frozenset()
 
@@ -317,9 +326,19 @@ class Trainer:
                 self.agg.start_row(self.payload, 2 * l + 1)
 
         t_comp = time.perf_counter()
+        ev_comp = self._event()
         grad = self.agg.aggregate(self.payload, step)
+        ev_agg = self._event()
         t_agg = time.perf_counter()
-        self.opt.step(grad)
+        skipped = False
+        if self.cfg.nan_guard and not bool(torch.isfinite(grad).all()):
+            # failure detection: never apply a non-finite decoded gradient
+            # (forces a device sync; the step syncs at the end anyway)
+            skipped = True
+            self.logger.log({"step": step, "event": "nan_grad_skipped"})
+        else:
+            self.opt.step(grad)
+        ev_end = self._event()
         if self.device.type == "cuda":
             torch.cuda.synchronize()
         t1 = time.perf_counter()
@@ -333,6 +352,14 @@ class Trainer:
             "agg": t_agg - t_comp,
             "update": t1 - t_agg,
         }
+        if skipped:
+            rec["skipped_update"] = True
+        if self._ev_step is not None and ev_end is not None:
+            # device-accurate spans (host timers only measure launch time with
+            # async collectives): step-start -> agg-start -> agg-end -> step-end
+            rec["gpu_comp"] = self._ev_step.elapsed_time(ev_comp) / 1e3
+            rec["gpu_agg"] = ev_comp.elapsed_time(ev_agg) / 1e3
+            rec["gpu_update"] = ev_agg.elapsed_time(ev_end) / 1e3
         if cfg.eval_freq > 0 and self.step_num % cfg.eval_freq == 0:
             if self.rank == 0:
                 self.save()
