@@ -1,0 +1,90 @@
+"""Convergence-under-attack experiment (the reference paper's evaluation methodology,
+SURVEY §4.4): plain averaging under a rev_grad adversary diverges; Draco's coded
+decodes track the clean curve.  Writes JSON + markdown to profiles/.
+
+  python tools/convergence.py --steps 200 --network ResNet18 --dataset Cifar10
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+
+
+def run_arm(name, steps, base, **kw):
+    import torch
+
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(**{**base, **kw})
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    losses = []
+    for _ in range(steps):
+        losses.append(t.train_step()["loss"])
+    acc = t.evaluate(n_batches=4)
+    t.close()
+    del t
+    if torch.cuda.is_available():
+        torch.cuda.empty_cache()
+    return {"name": name, "losses": losses, "prec1": acc["prec1"], "final_loss": losses[-1]}
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--network", type=str, default="ResNet18")
+    p.add_argument("--dataset", type=str, default="Cifar10")
+    p.add_argument("--batch-size", type=int, default=128)
+    p.add_argument("--lr", type=float, default=0.02)
+    p.add_argument("--out", type=str, default="profiles/convergence")
+    args = p.parse_args()
+
+    base = dict(network=args.network, dataset=args.dataset, batch_size=args.batch_size,
+                lr=args.lr, momentum=0.5, err_mode="rev_grad", max_steps=args.steps + 10,
+                eval_freq=0, log_dir="", train_dir="/tmp/conv_ckpt")
+
+    arms = [
+        run_arm("clean (mean, no adversary)", args.steps, base,
+                approach="baseline", mode="normal", worker_fail=0),
+        run_arm("attacked mean (s=1 rev_grad)", args.steps, base,
+                approach="baseline", mode="normal", worker_fail=1),
+        run_arm("Draco repetition r=3 (s=1 rev_grad)", args.steps, base,
+                approach="maj_vote", mode="maj_vote", group_size=3, worker_fail=1),
+        run_arm("Draco cyclic r=3 (s=1 rev_grad)", args.steps, base,
+                approach="cyclic", mode="cyclic", worker_fail=1, workers_per_rank=4),
+    ]
+
+    out = {"config": vars(args), "arms": [
+        {**a, "losses": [round(float(x), 5) for x in a["losses"]]} for a in arms
+    ]}
+    os.makedirs(os.path.dirname(args.out) or ".", exist_ok=True)
+    with open(args.out + ".json", "w") as f:
+        json.dump(out, f)
+
+    def sample(xs, k=10):
+        idx = np.linspace(0, len(xs) - 1, k).astype(int)
+        return " ".join(f"{xs[i]:.3f}" for i in idx)
+
+    with open(args.out + ".md", "w") as f:
+        f.write(f"# Convergence under attack — {args.network}/{args.dataset}, "
+                f"{args.steps} steps, lr={args.lr}, rev_grad adversary (seed-428 schedule)\n\n")
+        f.write("| arm | final loss | Prec@1 | loss trajectory (10 samples) |\n|---|---|---|---|\n")
+        for a in arms:
+            fl = a["final_loss"]
+            fl_s = f"{fl:.4f}" if np.isfinite(fl) else "diverged (NaN/inf)"
+            f.write(f"| {a['name']} | {fl_s} | {a['prec1']:.3f} | {sample(a['losses'])} |\n")
+        f.write("\nMethodology: single GPU hosts all logical workers (world=1); the\n"
+                "adversary replaces its gradient with -100x at the send boundary per the\n"
+                "deterministic seed-428 schedule, exactly as in the reference.\n")
+    print(json.dumps({a["name"]: (a["final_loss"], a["prec1"]) for a in arms}, indent=1))
+
+
+if __name__ == "__main__":
+    main()
