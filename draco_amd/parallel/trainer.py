@@ -126,8 +126,10 @@ class Trainer:
                                         comm_dtype=self._comm_dtype())
             self.data = GlobalBatchSource(self._dataset(), cfg.batch_size, n_workers=self.n)
             # encoded complex payload planes + raw sub-batch gradient scratch
+            # (one row per (worker, sub-batch) so ALL L*(2s+1) fwd/bwd graphs can
+            # replay concurrently; L*s_hat*d*4B — trivial against 288 GB HBM3E)
             self.payload = self.space.alloc_payload(self.L * 2)
-            self.scratch = self.space.alloc_payload(self.s_hat)
+            self.scratch = self.space.alloc_payload(self.L * self.s_hat)
             W = self.code.W
             sup = self.code.support
             self._w_re, self._w_im = [], []
@@ -142,6 +144,9 @@ class Trainer:
         self.use_graphs = bool(cfg.hip_graphs) and device.type == "cuda" and not cfg.deterministic
         self._graphs = {}
         self._streams = []
+        self.step_sync = True     # per-step device sync (bench may disable; driver
+                                  # brackets with its own barrier+synchronize)
+        self.collect_loss = True  # read losses to host each step
         self.n_fail = min(cfg.worker_fail, self.P)
         self.schedule = AdversarySchedule(self.P, self.n_fail, cfg.max_steps)
         self.step_num = 0
@@ -203,13 +208,14 @@ class Trainer:
     # writes would race across replicas, so only graph 0 updates them (the other
     # graphs are captured with BN momentum frozen to 0, making their stat writes
     # read-modify-write identities; stats are eval-only and rank-averaged anyway).
-    def _run_fwd_bwd(self, key, grad_row: torch.Tensor, x, y, stream=None):
+    def _run_fwd_bwd(self, key, grad_row: torch.Tensor, x, y, stream=None, freeze_bn=None):
         if not self.use_graphs:
             return self._forward_backward(x, y, grad_row)
         g = self._graphs.get(key)
         if g is None:
             try:
-                g = self._capture(grad_row, x, y, freeze_bn_stats=key[1] != 0)
+                g = self._capture(grad_row, x, y,
+                                  freeze_bn_stats=(key[1] != 0) if freeze_bn is None else freeze_bn)
             except RuntimeError as e:  # pragma: no cover - capture unsupported
                 import warnings
 
@@ -306,20 +312,28 @@ class Trainer:
                 # overlap: this row's all_to_all runs while other backwards compute
                 self.agg.start_row(self.payload, l)
         else:  # cyclic
+            # phase 1: every (worker, sub-batch) fwd/bwd replays concurrently
+            streams = {}
             for l in range(self.L):
                 w_global = l * self.world + self.rank
                 sup = self.code.support[w_global]
-                streams = []
                 for k in range(self.s_hat):
                     x, y = self.data.sub_batch(int(sup[k]), step)
-                    st = self._worker_stream(k)
-                    losses.append(self._run_fwd_bwd(("sub", k), self.scratch[k], x, y, stream=st))
-                    streams.append(st)
-                for st in streams:
+                    idx = l * self.s_hat + k
+                    st = self._worker_stream(idx)
+                    losses.append(self._run_fwd_bwd(("sub", l, k), self.scratch[idx], x, y,
+                                                    stream=st, freeze_bn=idx != 0))
+                    streams[idx] = st
+            # phase 2: join each worker's sub-batch streams, encode, inject, exchange
+            for l in range(self.L):
+                w_global = l * self.world + self.rank
+                for k in range(self.s_hat):
+                    st = streams[l * self.s_hat + k]
                     if st is not None:
                         torch.cuda.current_stream().wait_stream(st)
                 enc = self.payload[2 * l : 2 * l + 2]
-                ops.cyclic_encode(self.scratch, self._w_re[l], self._w_im[l], enc)
+                ops.cyclic_encode(self.scratch[l * self.s_hat : (l + 1) * self.s_hat],
+                                  self._w_re[l], self._w_im[l], enc)
                 if w_global in adversaries:
                     self._inject_encoded(enc, cfg.err_mode)
                 self.agg.start_row(self.payload, 2 * l)
@@ -339,14 +353,18 @@ class Trainer:
         else:
             self.opt.step(grad)
         ev_end = self._event()
-        if self.device.type == "cuda":
+        if self.device.type == "cuda" and self.step_sync:
             torch.cuda.synchronize()
         t1 = time.perf_counter()
 
         self.step_num += 1
+        if self.collect_loss:
+            loss_val = float(sum(float(v) for v in losses) / len(losses))
+        else:
+            loss_val = None  # skip the device sync of reading losses (bench hot loop)
         rec = {
             "step": step,
-            "loss": float(sum(float(v) for v in losses) / len(losses)),
+            "loss": loss_val,
             "time": t1 - t0,
             "comp": t_comp - t0,
             "agg": t_agg - t_comp,
