@@ -1,3 +1,11 @@
+from .real import PrefetchLoader, RealClassification, dataset_available
 from .synthetic import GlobalBatchSource, GroupBatchSource, SyntheticClassification
 
-__all__ = ["SyntheticClassification", "GroupBatchSource", "GlobalBatchSource"]
+__all__ = [
+    "SyntheticClassification",
+    "GroupBatchSource",
+    "GlobalBatchSource",
+    "RealClassification",
+    "PrefetchLoader",
+    "dataset_available",
+]

@@ -93,6 +93,10 @@ class _PSBase:
         self.schedule = AdversarySchedule(self.P, self.n_fail, cfg.max_steps)
 
     def _data(self):
+        from ..data.real import RealClassification, dataset_available
+
+        if dataset_available(self.cfg.dataset, self.cfg.data_root):
+            return RealClassification(self.cfg.dataset, self.cfg.data_root, self.device)
         return SyntheticClassification(self.cfg.dataset, self.device, seed=1234)
 
 

@@ -168,7 +168,11 @@ class Trainer:
         # "compress" = reference CLI alias (README.md:118, blosc) -> bf16 wire dtype
         return torch.bfloat16 if self.cfg.compress_grad in ("bf16", "compress") else torch.float32
 
-    def _dataset(self) -> SyntheticClassification:
+    def _dataset(self):
+        from ..data.real import RealClassification, dataset_available
+
+        if dataset_available(self.cfg.dataset, self.cfg.data_root):
+            return RealClassification(self.cfg.dataset, self.cfg.data_root, self.device)
         return SyntheticClassification(self.cfg.dataset, self.device, seed=1234)
 
     def _baseline_aggregator(self, cfg: Config):
