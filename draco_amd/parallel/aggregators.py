@@ -211,7 +211,9 @@ class GeoMedianAggregator(Aggregator):
         recv = self.exchanged(payload)  # (P, shard)
         P = recv.shape[0]
         z = recv.mean(dim=0)  # init at the mean (hdmedians does the same)
-        z_new = torch.empty_like(z)
+        # zeros, not empty: the pad tail past the last segment is never written by
+        # segment_weighted_mean, and garbage there (inf) poisons the delta criterion
+        z_new = torch.zeros_like(z)
         for _ in range(self.max_iter):
             part = ops.segment_sqdist(recv, z, self.local_seg)  # (P, L)
             self.comm.all_reduce(part)
@@ -224,7 +226,8 @@ class GeoMedianAggregator(Aggregator):
             scale = z_new.pow(2).sum()
             self.comm.all_reduce(scale)
             z, z_new = z_new, z
-            if float(delta) <= self.tol * self.tol * max(float(scale), 1e-12):
+            d_val, s_val = float(delta), float(scale)
+            if np.isfinite(d_val) and d_val <= self.tol * self.tol * max(s_val, 1e-12):
                 break
         self.comm.all_gather_shard(z, self._out)
         return self._out
