@@ -100,8 +100,18 @@ def main():
         elapsed = float(el_dev[0])
 
     ms_per_step = elapsed / args.steps * 1000.0
-    # distinct images per step: one group-batch per group, G = world groups
-    distinct_per_step = world * args.batch_size
+    # DISTINCT images per step (redundant compute is the price of the code and is
+    # not counted): maj_vote -> G*B (G=world groups); cyclic -> n*B global batch;
+    # baseline -> world*B
+    if args.approach == "cyclic":
+        distinct_per_step = t.n * args.batch_size
+        processed = t.n * (2 * args.worker_fail + 1) * args.batch_size
+    elif args.approach == "maj_vote":
+        distinct_per_step = world * args.batch_size
+        processed = world * args.group_size * args.batch_size
+    else:
+        distinct_per_step = world * args.batch_size
+        processed = distinct_per_step
     images_per_sec = distinct_per_step * args.steps / elapsed
 
     if rank == 0:
@@ -123,7 +133,7 @@ def main():
             "config": {
                 "model": args.network,
                 "global_batch": distinct_per_step,
-                "images_processed_per_step": world * args.group_size * args.batch_size,
+                "images_processed_per_step": processed,
                 "seq_len": None,
                 "input": "3x32x32" if args.dataset == "Cifar10" else args.dataset,
                 "parallelism": f"coded-dp{world}(repetition r={args.group_size}, s={args.worker_fail}, {args.err_mode})",
