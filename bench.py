@@ -37,7 +37,7 @@ def main():
     p.add_argument("--err-mode", type=str, default="rev_grad")
     p.add_argument("--dtype", type=str, default="bf16")
     p.add_argument("--device", type=str, default="auto")
-    p.add_argument("--channels-last", type=lambda v: v.lower() in ("1","true"), default=False)
+    p.add_argument("--channels-last", type=lambda v: v.lower() in ("1","true"), default=True)
     p.add_argument("--hip-graphs", type=lambda v: v.lower() in ("1","true"), default=True)
     args = p.parse_args()
 
