@@ -130,6 +130,7 @@ class Master(_PSBase):
         self.gather_buf = self.space.alloc_payload(self.P * self.payload_rows)
         self.logger = MetricsLogger(cfg.log_dir, self.rank)
         self.step_num = 0
+        self._abort = False
         if cfg.checkpoint_step > 0:
             load_checkpoint(
                 os.path.join(cfg.train_dir, f"model_step_{cfg.checkpoint_step}"),
@@ -137,12 +138,25 @@ class Master(_PSBase):
             )
             self.step_num = cfg.checkpoint_step
 
+    def request_abort(self):
+        """Preempt workers MID-STEP (the reference's tag-77 kill channel,
+        lenet.py:237-240 / resnet_split.py:636-640 — which had no master-side sender;
+        this is it).  Workers abandon remaining redundant sub-batches, send what they
+        have, and exit at the next step boundary."""
+        if self._abort:
+            return
+        self._abort = True
+        for w in range(self.P):
+            self.comm.isend(torch.zeros(1), dst=w + 1, tag=77)
+
     def run(self, max_steps: int | None = None):
         cfg = self.cfg
         steps = max_steps or cfg.max_steps
         ctrl = torch.zeros(2, dtype=torch.int64,
                            device=self.device if self.comm.backend == "nccl" else "cpu")
         for _ in range(steps):
+            if self._abort:
+                break
             t0 = time.perf_counter()
             # step announce (reference tag-10 broadcast, baseline_master.py:156-162) +
             # abort flag (the reference's vestigial tag-77 kill channel, done properly)
@@ -268,6 +282,29 @@ class Worker(_PSBase):
         self.payload = self.space.alloc_payload(self.payload_rows)
         self.logger = MetricsLogger(cfg.log_dir, self.rank)
         self.step_num = 0
+        self._preempt = self._arm_preempt()
+
+    def _arm_preempt(self):
+        """Out-of-band preemption: tag-77 irecv from the master, observed through a
+        daemon waiter thread (gloo Work.is_completed never fires for p2p recvs
+        without wait())."""
+        import threading
+
+        ev = threading.Event()
+        buf = torch.zeros(1)
+        try:
+            req = self.comm.irecv(buf, src=0, tag=77)
+        except Exception:
+            return ev  # backend without tagged p2p: preemption disabled
+
+        def waiter():
+            try:
+                req.wait()
+            finally:
+                ev.set()
+
+        threading.Thread(target=waiter, daemon=True).start()
+        return ev
 
     def run(self, max_steps: int | None = None):
         cfg = self.cfg
@@ -286,12 +323,14 @@ class Worker(_PSBase):
                 sup = self.code.support[self.worker_id]
                 losses = []
                 for k in range(self.code.s_hat):
+                    if self._preempt.is_set():
+                        break  # preempted: ship what we have, exit at next ctrl
                     x, y = self.data.sub_batch(int(sup[k]), self.step_num)
                     losses.append(self._fwd_bwd(x, y, self.scratch[k]))
                 ops.cyclic_encode(self.scratch, self._w_re, self._w_im, self.payload)
                 if self.worker_id in adversaries:
                     _inject_encoded(self.payload, cfg.err_mode)
-                loss = float(np.mean(losses))
+                loss = float(np.mean(losses)) if losses else float("nan")
             else:
                 x, y = self.data.batch_for(self.group, self.step_num)
                 loss = self._fwd_bwd(x, y, self.payload[0])

@@ -76,3 +76,39 @@ def test_ps_straggler_becomes_erasure():
     # 1 PS + 4 cyclic workers (s=1); worker 2 stalls on step 1 -> erased, training continues
     res = run_dist(_ps_straggler_worker, 5, 2, 3, timeout=240)
     assert np.isfinite(res[0]) and res[0] > 0
+
+
+def _ps_abort_worker(rank, world):
+    """Master preempts mid-run after step 1; everyone exits long before max_steps."""
+    import time as _t
+
+    from draco_amd.config import Config
+    from draco_amd.parallel.ps import Master, Worker
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                 topology="ps", approach="cyclic", mode="cyclic", worker_fail=1,
+                 err_mode="none", max_steps=500, eval_freq=0, log_dir="",
+                 train_dir="/tmp/draco_abort")
+    t0 = _t.time()
+    if rank == 0:
+        m = Master(cfg)
+        orig = m._gather_grads
+
+        def gg():
+            e = orig()
+            if m.step_num == 1:
+                m.request_abort()
+            return e
+
+        m._gather_grads = gg
+        m.run()  # max_steps=500; abort must cut it short
+        return _t.time() - t0
+    w = Worker(cfg)
+    w.run()
+    return _t.time() - t0
+
+
+def test_ps_master_abort_preempts():
+    res = run_dist(_ps_abort_worker, 5, timeout=180)
+    # 500 steps would take minutes; abort after step 1 ends the run in seconds
+    assert max(res.values()) < 90
