@@ -39,6 +39,7 @@ def main():
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--channels-last", type=lambda v: v.lower() in ("1","true"), default=True)
     p.add_argument("--hip-graphs", type=lambda v: v.lower() in ("1","true"), default=True)
+    p.add_argument("--compile", type=lambda v: v.lower() in ("1","true"), default=False)
     args = p.parse_args()
 
     from draco_amd.config import Config
@@ -66,6 +67,7 @@ def main():
         topology="colocated",
         channels_last=args.channels_last,
         hip_graphs=args.hip_graphs,
+        compile=args.compile,
     )
     t = Trainer(cfg)
     t.logger.stdout_every = 0

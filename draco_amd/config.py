@@ -51,6 +51,7 @@ class Config:
                                     # missing workers become erasures (0 = wait forever)
     nan_guard: bool = True        # failure detection: skip updates on non-finite decode
     gpu_timing: bool = False      # device-accurate phase spans via HIP events (metrics)
+    compile: bool = False         # torch.compile the model before hipGraph capture
     data_root: str = ""           # directory with real MNIST/CIFAR files (IDX/pickle);
                                   # empty or missing files -> deterministic synthetic data
     log_dir: str = "output/logs/"

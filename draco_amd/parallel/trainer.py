@@ -73,6 +73,8 @@ class Trainer:
         torch.manual_seed(cfg.seed)
         self.model = build_model(cfg.network, cfg.dataset).to(device)
         self.model.train()
+        if cfg.compile and device.type == "cuda":
+            self.model = torch.compile(self.model)
         self.use_cl = cfg.channels_last and device.type == "cuda"
         self.space = FlatSpace(self.model, self.world, device, channels_last=self.use_cl)
         if cfg.optimizer == "adam":
