@@ -39,7 +39,7 @@ def main():
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--channels-last", type=lambda v: v.lower() in ("1","true"), default=True)
     p.add_argument("--hip-graphs", type=lambda v: v.lower() in ("1","true"), default=True)
-    p.add_argument("--compile", type=lambda v: v.lower() in ("1","true"), default=False)
+    p.add_argument("--compile", type=lambda v: v.lower() in ("1","true"), default=True)
     args = p.parse_args()
 
     from draco_amd.config import Config
