@@ -74,9 +74,9 @@ class Trainer:
         self.model = build_model(cfg.network, cfg.dataset).to(device)
         self.model.train()
         if cfg.compile and device.type == "cuda":
-            import torch._dynamo
+            import torch._dynamo as _dynamo
 
-            torch._dynamo.config.suppress_errors = True  # fall back to eager per-op
+            _dynamo.config.suppress_errors = True  # fall back to eager per-op
             self.model = torch.compile(self.model)
         self.use_cl = cfg.channels_last and device.type == "cuda"
         self.space = FlatSpace(self.model, self.world, device, channels_last=self.use_cl)
