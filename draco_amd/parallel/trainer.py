@@ -410,19 +410,24 @@ class Trainer:
     # ------------------------------------------------------------------ eval / io
     @torch.no_grad()
     def evaluate(self, n_batches: int = 8) -> dict:
+        """Held-out Prec@1/Prec@5 + loss (reference eval cadence,
+        baseline_worker.py:148-155 / distributed_evaluator.py:92-110)."""
         self.model.eval()
         data = self._dataset()
-        correct = total = 0
+        correct = correct5 = total = 0
         loss_sum = 0.0
         for b in range(n_batches):
-            # held-out stream: negative index space never touched by training
+            # held-out stream: index space never touched by training
             x, y = data.get_batch(2**40 + b * self.cfg.test_batch_size, self.cfg.test_batch_size)
             logits = self.model(x)
             loss_sum += float(self.criterion(logits, y))
-            correct += int((logits.argmax(dim=1) == y).sum())
+            k = min(5, logits.shape[1])
+            topk = logits.topk(k, dim=1).indices
+            correct += int((topk[:, 0] == y).sum())
+            correct5 += int((topk == y[:, None]).any(dim=1).sum())
             total += y.numel()
         self.model.train()
-        return {"prec1": correct / total, "loss": loss_sum / n_batches}
+        return {"prec1": correct / total, "prec5": correct5 / total, "loss": loss_sum / n_batches}
 
     def _ckpt_path(self, step: int | None = None) -> str:
         step = self.step_num if step is None else step

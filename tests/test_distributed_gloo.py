@@ -216,3 +216,47 @@ def _compress_worker(rank, world):
 
 def test_bf16_wire_compression():
     run_dist(_compress_worker, 2)
+
+
+def _tolerance_vote_worker(rank, world):
+    """Distributed tolerance vote: honest members differ by sub-threshold noise on
+    DIFFERENT shards; the adversary is excluded; the cross-shard MAX-allreduce makes
+    every rank pick the same winner."""
+    import torch.nn as nn
+
+    from draco_amd.parallel.aggregators import VoteAggregator
+    from draco_amd.parallel.flat import FlatSpace
+
+    comm = _comm(rank, world)
+    torch.manual_seed(7)
+    model = nn.Linear(50, 10)
+    space = FlatSpace(model, world, torch.device("cpu"))
+    r = 3
+    agg = VoteAggregator(comm, space, group_size=r, atol=0.0, rtol=0.1)
+    payload = space.alloc_payload(r)
+    for l in range(r):
+        g = (rank - l) % world
+        torch.manual_seed(1000 + g)
+        honest = torch.randn(space.d_pad)
+        if l == g % r:
+            payload[l] = honest * -100.0  # adversary: way outside the tolerance ball
+        else:
+            # sub-threshold replica noise, concentrated on THIS rank's shard only
+            noise = torch.zeros(space.d_pad)
+            lo = rank * space.shard
+            noise[lo : lo + space.shard] = 0.001
+            payload[l] = honest + noise * (1 if l == 0 else -1)
+    out = agg.aggregate(payload, step=0)
+    ref = torch.zeros(space.d_pad)
+    for g in range(world):
+        torch.manual_seed(1000 + g)
+        ref += torch.randn(space.d_pad)
+    ref /= world
+    # winner is an honest member (within the small injected noise)
+    assert float((out - ref).abs().max()) < 0.01, float((out - ref).abs().max())
+    comm.shutdown()
+    return True
+
+
+def test_distributed_tolerance_vote():
+    run_dist(_tolerance_vote_worker, 3)

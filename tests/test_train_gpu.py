@@ -101,6 +101,19 @@ def test_gpu_vote_excludes_adversary(tmp_path):
     t.close()
 
 
+def test_gpu_compiled_train(tmp_path):
+    """torch.compile path (the bench default) trains and matches eager loss scale."""
+    from draco_amd.parallel.trainer import Trainer
+
+    t = Trainer(_cfg(tmp_path, approach="maj_vote", mode="maj_vote", group_size=3,
+                     worker_fail=1, lr=0.02, compile=True))
+    t.logger.stdout_every = 0
+    losses = [t.train_step()["loss"] for _ in range(8)]
+    assert np.isfinite(losses).all()
+    assert min(losses[-3:]) < losses[0] * 2.0
+    t.close()
+
+
 def test_gpu_checkpoint_roundtrip(tmp_path):
     from draco_amd.parallel.trainer import Trainer
 
