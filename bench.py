@@ -101,6 +101,10 @@ def main():
         t.comm.all_reduce(el_dev, op="max")
         elapsed = float(el_dev[0])
 
+    # post-bracket honesty telemetry: one more step with loss readback
+    t.collect_loss = True
+    t.step_sync = True
+    final_loss = t.train_step()["loss"]
     ms_per_step = elapsed / args.steps * 1000.0
     # DISTINCT images per step (redundant compute is the price of the code and is
     # not counted): maj_vote -> G*B (G=world groups); cyclic -> n*B global batch;
@@ -130,6 +134,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
+            "final_loss": round(final_loss, 4) if final_loss == final_loss else None,
+            "skipped_updates": t.skipped_updates,
             "dtype": args.dtype if on_gpu else "fp32",
             "data": "synthetic",
             "config": {

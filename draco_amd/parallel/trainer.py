@@ -155,6 +155,7 @@ class Trainer:
         self.n_fail = min(cfg.worker_fail, self.P)
         self.schedule = AdversarySchedule(self.P, self.n_fail, cfg.max_steps)
         self.step_num = 0
+        self.skipped_updates = 0
         self.logger = MetricsLogger(cfg.log_dir, self.rank)
         self.criterion = F.cross_entropy
 
@@ -214,17 +215,20 @@ class Trainer:
     # (private mempools, disjoint payload rows, shared READ-ONLY params), and a
     # 128-image CIFAR conv underfills 256 CUs — so each worker's graph replays on its
     # own HIP stream and the three fwd+bwd overlap on the chip.  BN running-stat
-    # writes would race across replicas, so only graph 0 updates them (the other
-    # graphs are captured with BN momentum frozen to 0, making their stat writes
-    # read-modify-write identities; stats are eval-only and rank-averaged anyway).
+    # updates race across concurrent replicas; the race is benign (stats are
+    # eval-only, approximately equal across replicas, and rank-averaged before
+    # checkpoints).  NOTE: do NOT "fix" the race by freezing BN momentum in replica
+    # graphs — under torch.compile a momentum change triggers a RECOMPILE, so the
+    # replicas would run different kernels, their gradients would diverge beyond the
+    # vote tolerance, and the majority vote breaks (observed: adversary winning 1/3
+    # of votes).  Identical code across replicas is what the vote relies on.
     def _run_fwd_bwd(self, key, grad_row: torch.Tensor, x, y, stream=None, freeze_bn=None):
         if not self.use_graphs:
             return self._forward_backward(x, y, grad_row)
         g = self._graphs.get(key)
         if g is None:
             try:
-                g = self._capture(grad_row, x, y,
-                                  freeze_bn_stats=(key[1] != 0) if freeze_bn is None else freeze_bn)
+                g = self._capture(grad_row, x, y)
             except RuntimeError as e:  # pragma: no cover - capture unsupported
                 import warnings
 
@@ -251,7 +255,7 @@ class Trainer:
             self._streams.append(torch.cuda.Stream())
         return self._streams[idx]
 
-    def _capture(self, grad_row: torch.Tensor, x, y, freeze_bn_stats: bool = False):
+    def _capture(self, grad_row: torch.Tensor, x, y):
         self.space.attach_grads(grad_row)
         static_x = x.clone()
         static_y = y.clone()
@@ -266,29 +270,19 @@ class Trainer:
             loss.backward()
             return loss
 
-        bn_mom = []
-        if freeze_bn_stats:
-            for m in self.model.modules():
-                if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
-                    bn_mom.append((m, m.momentum))
-                    m.momentum = 0.0  # frozen into the captured BN kernels
-        try:
-            # warm up on a side stream (MIOpen find, autograd graph materialisation)
-            s = torch.cuda.Stream()
-            s.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(s):
-                for _ in range(2):
-                    body()
-            torch.cuda.current_stream().wait_stream(s)
+        # warm up on a side stream (MIOpen find, dynamo compile, autograd warmup)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                body()
+        torch.cuda.current_stream().wait_stream(s)
 
-            # private mempool per graph: graphs sharing a pool may alias activation
-            # memory, which forbids concurrent replay
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                loss = body()
-        finally:
-            for m, mom in bn_mom:
-                m.momentum = mom
+        # private mempool per graph: graphs sharing a pool may alias activation
+        # memory, which forbids concurrent replay
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            loss = body()
         return {"graph": graph, "x": static_x, "y": static_y, "loss": loss.detach()}
 
     # ------------------------------------------------------------------ one step
@@ -331,7 +325,7 @@ class Trainer:
                     idx = l * self.s_hat + k
                     st = self._worker_stream(idx)
                     losses.append(self._run_fwd_bwd(("sub", l, k), self.scratch[idx], x, y,
-                                                    stream=st, freeze_bn=idx != 0))
+                                                    stream=st))
                     streams[idx] = st
             # phase 2: join each worker's sub-batch streams, encode, inject, exchange
             for l in range(self.L):
@@ -358,6 +352,7 @@ class Trainer:
             # failure detection: never apply a non-finite decoded gradient
             # (forces a device sync; the step syncs at the end anyway)
             skipped = True
+            self.skipped_updates += 1
             self.logger.log({"step": step, "event": "nan_grad_skipped"})
         else:
             self.opt.step(grad)
