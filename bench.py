@@ -19,6 +19,15 @@ import os
 import sys
 import time
 
+# Cross-rank kernel determinism: repetition-vote group members live on DIFFERENT
+# ranks, so every rank must run the SAME conv/compiled kernels or replica gradients
+# diverge systematically (beyond fp-reorder noise).  FAST find picks MIOpen algos
+# from the heuristic (deterministic across ranks; measured perf-neutral vs full
+# find), and a shared inductor cache makes all ranks reuse identical compiled
+# kernels.  Must be set before torch/MIOpen initialise.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+os.environ.setdefault("TORCHINDUCTOR_CACHE_DIR", "/tmp/draco_inductor_cache")
+
 import torch
 
 
