@@ -145,6 +145,7 @@ def main():
             "vs_baseline": None,
             "final_loss": round(final_loss, 4) if final_loss == final_loss else None,
             "skipped_updates": t.skipped_updates,
+            "vote_degenerate_steps": getattr(t.agg, "degenerate_steps", None),
             "dtype": args.dtype if on_gpu else "fp32",
             "data": "synthetic",
             "config": {

@@ -155,6 +155,10 @@ class VoteAggregator(Aggregator):
         self.pairs_a = torch.tensor(pairs_a, dtype=torch.int64, device=space.device)
         self.pairs_b = torch.tensor(pairs_b, dtype=torch.int64, device=space.device)
         self.n_pairs_per_group = self.r * (self.r - 1) // 2
+        # telemetry: steps where some group saw NO equal pair (tolerance too tight /
+        # replicas diverged -> Boyer-Moore degenerates to "last member wins", which an
+        # adversary can exploit).  Surfaced by bench as vote_degenerate_steps.
+        self.degenerate_steps = 0
 
     @classmethod
     def from_member_rows(cls, comm, space, member_rows, atol: float = 0.0, rtol: float = 0.0):
@@ -177,13 +181,20 @@ class VoteAggregator(Aggregator):
         eq_host = eq.to("cpu", non_blocking=False).numpy()
         winners = np.empty(self.G, dtype=np.int64)
         k = 0
+        degenerate = False
         for g in range(self.G):
             mat = np.eye(self.r, dtype=bool)
+            any_eq = False
             for i in range(self.r):
                 for j in range(i + 1, self.r):
                     mat[i, j] = mat[j, i] = bool(eq_host[k])
+                    any_eq = any_eq or bool(eq_host[k])
                     k += 1
+            if not any_eq and self.r > 1:
+                degenerate = True
             winners[g] = self.member_rows[g, majority_vote_index(mat)]
+        if degenerate:
+            self.degenerate_steps += 1
         idx = torch.tensor(winners, dtype=torch.int64, device=recv.device)
         ops.mean_rows(recv, idx, self._shard_out)
         self.comm.all_gather_shard(self._shard_out, self._out)

@@ -170,6 +170,9 @@ class Master(_PSBase):
                 grad = self.agg.aggregate(self.gather_buf, self.step_num, erasures=erasures)
             else:
                 grad = self.agg.aggregate(self.gather_buf, self.step_num)
+                if erasures and cfg.approach == "baseline" and cfg.mode == "normal":
+                    # missing rows were zeroed; rescale the mean to stay unbiased
+                    grad = grad * (self.P / max(self.P - len(erasures), 1))
             t_agg = time.perf_counter()
             self.opt.step(grad)
             t1 = time.perf_counter()
