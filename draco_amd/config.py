@@ -5,7 +5,7 @@ extensions (topology, dtype, workers-per-rank, vote-atol).
 from __future__ import annotations
 
 import argparse
-from dataclasses import dataclass, field, fields
+from dataclasses import dataclass, fields
 
 
 @dataclass

@@ -5,7 +5,6 @@ lenet.py:43-341) is intentionally NOT reproduced: the framework gets the same
 comm/compute overlap from flat gradient views + stream-ordered collectives
 (draco_amd/parallel/flat.py), so one plain autograd module serves both roles.
 """
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

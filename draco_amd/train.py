@@ -12,9 +12,6 @@ Topologies:
 """
 from __future__ import annotations
 
-import os
-import sys
-
 
 def main(argv=None):
     from .config import parse_cli
