@@ -120,7 +120,9 @@ def main():
     # baseline -> world*B
     if args.approach == "cyclic":
         distinct_per_step = t.n * args.batch_size
-        processed = t.n * (2 * args.worker_fail + 1) * args.batch_size
+        # actually computed per step: each rank computes its DISTINCT local
+        # sub-batches once (band overlap between same-rank logical workers deduped)
+        processed = world * len(t._local_subs) * args.batch_size
     elif args.approach == "maj_vote":
         distinct_per_step = world * args.batch_size
         processed = world * args.group_size * args.batch_size

@@ -130,19 +130,28 @@ class Trainer:
             self.agg = CyclicAggregator(self.comm, self.space, self.code, self.L,
                                         comm_dtype=self._comm_dtype())
             self.data = GlobalBatchSource(self._dataset(), cfg.batch_size, n_workers=self.n)
-            # encoded complex payload planes + raw sub-batch gradient scratch
-            # (one row per (worker, sub-batch) so ALL L*(2s+1) fwd/bwd graphs can
-            # replay concurrently; L*s_hat*d*4B — trivial against 288 GB HBM3E)
             self.payload = self.space.alloc_payload(self.L * 2)
-            self.scratch = self.space.alloc_payload(self.L * self.s_hat)
             W = self.code.W
             sup = self.code.support
-            self._w_re, self._w_im = [], []
-            for l in range(self.L):
-                w_global = l * self.world + self.rank
+            # Sub-batch dedup within a rank: logical workers hosted on the SAME rank
+            # have overlapping cyclic bands; their shared sub-batch gradients are
+            # computed ONCE (the reference recomputes per worker because its workers
+            # are separate machines — same-host recompute adds no fault isolation,
+            # and all transmitted encodings are identical either way).  At N=1 this
+            # is n fwd/bwd per step instead of n*(2s+1); at N=world=n (L=1) it is
+            # the usual 2s+1.
+            local_w = [l * self.world + self.rank for l in range(self.L)]
+            needed = sorted({int(j) for w in local_w for j in sup[w]})
+            self._local_subs = needed
+            row_of = {j: i for i, j in enumerate(needed)}
+            self.scratch = self.space.alloc_payload(len(needed))
+            self._w_re, self._w_im, self._enc_rows = [], [], []
+            for l, w_global in enumerate(local_w):
                 coeff = W[w_global, sup[w_global]]
                 self._w_re.append(torch.tensor(np.real(coeff), dtype=torch.float32, device=device))
                 self._w_im.append(torch.tensor(np.imag(coeff), dtype=torch.float32, device=device))
+                self._enc_rows.append(torch.tensor([row_of[int(j)] for j in sup[w_global]],
+                                                   dtype=torch.int64, device=device))
         else:
             raise ValueError(f"unknown approach {approach!r}")
 
@@ -315,28 +324,23 @@ class Trainer:
                 # overlap: this row's all_to_all runs while other backwards compute
                 self.agg.start_row(self.payload, l)
         else:  # cyclic
-            # phase 1: every (worker, sub-batch) fwd/bwd replays concurrently
-            streams = {}
+            # phase 1: every DISTINCT local sub-batch fwd/bwd replays concurrently
+            streams = []
+            for i, j in enumerate(self._local_subs):
+                x, y = self.data.sub_batch(j, step)
+                st = self._worker_stream(i)
+                losses.append(self._run_fwd_bwd(("sub", i), self.scratch[i], x, y, stream=st))
+                streams.append(st)
+            for st in streams:
+                if st is not None:
+                    torch.cuda.current_stream().wait_stream(st)
+            # phase 2: per logical worker, encode its band (gathered rows), inject,
+            # start the exchange
             for l in range(self.L):
                 w_global = l * self.world + self.rank
-                sup = self.code.support[w_global]
-                for k in range(self.s_hat):
-                    x, y = self.data.sub_batch(int(sup[k]), step)
-                    idx = l * self.s_hat + k
-                    st = self._worker_stream(idx)
-                    losses.append(self._run_fwd_bwd(("sub", l, k), self.scratch[idx], x, y,
-                                                    stream=st))
-                    streams[idx] = st
-            # phase 2: join each worker's sub-batch streams, encode, inject, exchange
-            for l in range(self.L):
-                w_global = l * self.world + self.rank
-                for k in range(self.s_hat):
-                    st = streams[l * self.s_hat + k]
-                    if st is not None:
-                        torch.cuda.current_stream().wait_stream(st)
                 enc = self.payload[2 * l : 2 * l + 2]
-                ops.cyclic_encode(self.scratch[l * self.s_hat : (l + 1) * self.s_hat],
-                                  self._w_re[l], self._w_im[l], enc)
+                ops.combine_rows(self.scratch, self._enc_rows[l], self._w_re[l], enc[0])
+                ops.combine_rows(self.scratch, self._enc_rows[l], self._w_im[l], enc[1])
                 if w_global in adversaries:
                     self._inject_encoded(enc, cfg.err_mode)
                 self.agg.start_row(self.payload, 2 * l)
