@@ -317,15 +317,20 @@ class CyclicAggregator(Aggregator):
         l, src = w_ids // world, w_ids % world
         self.rows_re = (2 * l) * world + src
         self.rows_im = (2 * l + 1) * world + src
+        self._z = None
+        self._zgen = None
 
     def aggregate(self, payload_planes: torch.Tensor, step: int,
                   erasures: frozenset = frozenset()) -> torch.Tensor:
         recv = self.exchanged(payload_planes)  # (2L*world, shard)
-        gen = torch.Generator(device="cpu")
-        gen.manual_seed(0x5EED ^ (step * 1000003) ^ self.comm.rank)
-        z = torch.normal(
-            mean=1.0, std=1.0, size=(self.space.shard,), generator=gen, dtype=torch.float32
-        ).to(self.space.device)
+        # random projection generated ON DEVICE (a shard-sized CPU randn + H2D copy
+        # costs ~30 ms at d=11M — it dominated the whole decode); per-rank streams
+        # may legitimately use different z (partial projections are summed)
+        if self._z is None:
+            self._z = torch.empty(self.space.shard, dtype=torch.float32, device=self.space.device)
+            self._zgen = torch.Generator(device=self.space.device)
+        self._zgen.manual_seed(0x5EED ^ (step * 1000003) ^ self.comm.rank)
+        z = self._z.normal_(mean=1.0, std=1.0, generator=self._zgen)
         proj = ops.cyclic_project(recv, z)  # (2L*world,) per-row partial dots
         self.comm.all_reduce(proj)
         pa = proj.to("cpu").numpy().astype(np.float64)
