@@ -32,6 +32,13 @@ import torch
 
 
 def main():
+    # The driver parses EXACTLY ONE JSON line from stdout.  Native libraries chat on
+    # fd 1 (e.g. LAPACK XERBLA "** On entry to ZLASCL ..."), so redirect fd 1 to
+    # stderr for the whole run and restore it only for the final JSON print.
+    real_stdout = os.dup(1)
+    os.dup2(2, 1)
+    sys.stdout = os.fdopen(os.dup(2), "w")
+
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
@@ -131,6 +138,10 @@ def main():
         processed = distinct_per_step
     images_per_sec = distinct_per_step * args.steps / elapsed
 
+    # restore the real stdout for the contract line
+    sys.stdout.flush()
+    os.dup2(real_stdout, 1)
+    sys.stdout = os.fdopen(real_stdout, "w")
     if rank == 0:
         result = {
             "metric": (f"images/sec (effective), {args.network} {args.dataset}, "
