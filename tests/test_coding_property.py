@@ -1,0 +1,72 @@
+"""Property-based coding tests (hypothesis): encode->decode exactness over random
+(n, s, adversary-set, error-pattern) draws."""
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from draco_amd.coding import build_cyclic_code, majority_vote_index
+
+_CODES = {}
+
+
+def _code(n, s):
+    if (n, s) not in _CODES:
+        _CODES[(n, s)] = build_cyclic_code(n, s)
+    return _CODES[(n, s)]
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    ns=st.sampled_from([(4, 1), (6, 1), (6, 2), (8, 2), (10, 3)]),
+    seed=st.integers(0, 2**31 - 1),
+    err_scale=st.sampled_from([1e-3, 1.0, 1e4]),
+)
+def test_cyclic_roundtrip_random(ns, seed, err_scale):
+    n, s = ns
+    code = _code(n, s)
+    rng = np.random.default_rng(seed)
+    d = 64
+    G = rng.normal(size=(n, d))
+    R = np.stack([code.encode_oracle(w, G[code.support[w]]) for w in range(n)])
+    nbad = int(rng.integers(0, s + 1))
+    bad = rng.choice(n, size=nbad, replace=False)
+    R[bad] += err_scale * (rng.normal(size=(nbad, d)) + 1j * rng.normal(size=(nbad, d)))
+    dec = code.decode_oracle(R, rng)
+    ref = G.sum(axis=0)
+    assert np.abs(dec - ref).max() < 1e-5 * max(np.abs(ref).max(), 1.0)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    r=st.sampled_from([3, 5, 7]),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_majority_vote_honest_majority_wins(r, seed):
+    """With a strict honest majority of IDENTICAL members, a corrupt member can
+    never win, regardless of order."""
+    rng = np.random.default_rng(seed)
+    n_bad = int(rng.integers(0, (r - 1) // 2 + 1))
+    labels = np.array([0] * (r - n_bad) + list(range(1, n_bad + 1)))
+    rng.shuffle(labels)
+    eq = labels[:, None] == labels[None, :]
+    w = majority_vote_index(eq)
+    assert labels[w] == 0
+
+
+@settings(max_examples=20, deadline=None)
+@given(seed=st.integers(0, 2**31 - 1))
+def test_cyclic_erasure_plus_error(seed):
+    code = _code(8, 2)
+    rng = np.random.default_rng(seed)
+    d = 48
+    G = rng.normal(size=(8, d))
+    R = np.stack([code.encode_oracle(w, G[code.support[w]]) for w in range(8)])
+    erase, err = rng.choice(8, size=2, replace=False)
+    R[erase] = 0.0
+    R[err] *= -99.0
+    z = rng.normal(loc=1.0, size=d)
+    healthy = code.locate_errors(code.W_perp @ (R @ z), known_bad={int(erase)})
+    v = code.recombination_vector(healthy)
+    dec = np.real(v @ R)
+    ref = G.sum(axis=0)
+    assert np.abs(dec - ref).max() < 1e-6 * max(np.abs(ref).max(), 1.0)

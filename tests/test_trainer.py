@@ -131,3 +131,24 @@ def test_evaluator_checkpoint(tmp_path):
     assert os.path.exists(path)
     rec = evaluate_checkpoint(path, torch.device("cpu"), batches=2)
     assert rec["step"] == 4 and 0.0 <= rec["prec1"] <= 1.0 and 0.0 <= rec["prec5"] <= 1.0
+
+
+def test_evaluator_polling_once(tmp_path):
+    """evaluate.py --once consumes checkpoints like the reference's polling process."""
+    import subprocess
+    import sys as _sys
+
+    cfg = _cfg(tmp_path, approach="baseline", mode="normal", worker_fail=0, eval_freq=3)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    for _ in range(6):
+        t.train_step()
+    t.close()
+    out = subprocess.run(
+        [_sys.executable, "-m", "draco_amd.evaluate", "--model-dir", cfg.train_dir,
+         "--once", "--eval-batch-size", "16"],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert "Testset Performance: Cur Step:3" in out.stdout
+    assert "Testset Performance: Cur Step:6" in out.stdout
