@@ -146,8 +146,8 @@ class Master(_PSBase):
         if self._abort:
             return
         self._abort = True
-        for w in range(self.P):
-            self.comm.isend(torch.zeros(1), dst=w + 1, tag=77)
+        self._abort_reqs = [self.comm.isend(torch.zeros(1), dst=w + 1, tag=77)
+                            for w in range(self.P)]
 
     def run(self, max_steps: int | None = None):
         cfg = self.cfg
@@ -192,6 +192,12 @@ class Master(_PSBase):
         ctrl[0] = self.step_num
         ctrl[1] = 1
         self.comm.broadcast(ctrl, src=0)
+        # complete every worker's armed tag-77 preemption irecv — a pending p2p recv
+        # makes the gloo context abort at process teardown
+        if not self._abort:
+            self.request_abort()
+        for r in getattr(self, "_abort_reqs", []):
+            r.wait()
         self.logger.close()
 
     def _gather_grads(self) -> frozenset:
@@ -306,7 +312,9 @@ class Worker(_PSBase):
             finally:
                 ev.set()
 
-        threading.Thread(target=waiter, daemon=True).start()
+        th = threading.Thread(target=waiter, daemon=True)
+        th.start()
+        self._preempt_thread = th
         return ev
 
     def run(self, max_steps: int | None = None):
@@ -346,6 +354,11 @@ class Worker(_PSBase):
                 "step": self.step_num, "role": "worker", "loss": loss,
                 "comp": t_comp - t0, "comm": t1 - t_comp, "time": t1 - t0,
             })
+        # the master completes the tag-77 irecv at shutdown; join the waiter so the
+        # recv is fully consumed before the process tears gloo down
+        th = getattr(self, "_preempt_thread", None)
+        if th is not None:
+            th.join(timeout=30)
         self.logger.close()
 
     def _fwd_bwd(self, x, y, row):
