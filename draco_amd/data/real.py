@@ -153,6 +153,10 @@ class PrefetchLoader:
         if self._next_start == start:
             if self._stream is not None:
                 torch.cuda.current_stream().wait_stream(self._stream)
+                # staged tensors were allocated on the side stream; record their use
+                # on the consumer stream so the allocator cannot recycle them early
+                for t in self._next:
+                    t.record_stream(torch.cuda.current_stream())
             out = self._next
         else:
             out = self.source.get_batch(start, self.B)

@@ -250,6 +250,13 @@ class Trainer:
             with torch.cuda.stream(stream):
                 g["x"].copy_(x)
                 g["y"].copy_(y)
+                # x/y were allocated on the DEFAULT stream and their refs die when
+                # this call returns; the caching allocator does not track cross-
+                # stream consumers, so without record_stream their blocks can be
+                # recycled and overwritten while the side-stream copy is still in
+                # flight (intermittent garbage batches -> inf/NaN gradients)
+                x.record_stream(stream)
+                y.record_stream(stream)
                 g["graph"].replay()
         else:
             g["x"].copy_(x)
