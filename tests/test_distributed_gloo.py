@@ -260,3 +260,25 @@ def _tolerance_vote_worker(rank, world):
 
 def test_distributed_tolerance_vote():
     run_dist(_tolerance_vote_worker, 3)
+
+
+def _adam_worker(rank, world):
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.005,
+                 approach="maj_vote", mode="maj_vote", group_size=3, worker_fail=1,
+                 optimizer="adam", max_steps=50, eval_freq=0, log_dir="",
+                 train_dir="/tmp/draco_adam")
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    losses = [t.train_step()["loss"] for _ in range(8)]
+    h = float(t.space.flat_param.double().sum())
+    t.close()
+    return (losses[0], losses[-1], h)
+
+
+def test_adam_distributed():
+    res = run_dist(_adam_worker, 2)
+    assert res[0][2] == res[1][2], "params diverged across ranks under Adam"
+    assert res[0][1] < res[0][0]
