@@ -28,6 +28,11 @@ def main(argv=None):
     try:
         for _ in range(cfg.max_steps):
             t.train_step()
+            if cfg.eval_freq > 0 and t.step_num % cfg.eval_freq == 0 and t.rank == 0:
+                m = t.evaluate()
+                print(f"Testset Performance: Cur Step:{t.step_num} "
+                      f"Prec@1: {m['prec1']:.4f} Prec@5: {m['prec5']:.4f} Loss: {m['loss']:.4f}",
+                      flush=True)
     finally:
         t.close()
 
