@@ -1,7 +1,13 @@
 """Checkpointing in the reference's file layout (`train_dir/model_step_N`,
 baseline_master.py:237-248), improved per SURVEY §5.4: state_dict + optimizer state
 (the reference drops momentum) written atomically (tmp + rename); `--checkpoint-step`
-resume supported on every rank."""
+resume supported on every rank.
+
+Layout independence: model weights restore through load_state_dict (per-parameter),
+and flat optimizer buffers (momentum / Adam moments) are stored PER PARAMETER via the
+flat-space views — so a checkpoint written under one flat layout (e.g.
+channels_last=False) resumes correctly under another.
+"""
 from __future__ import annotations
 
 import os
@@ -9,13 +15,52 @@ import os
 import torch
 
 
+def _flat_to_params(space, flat: torch.Tensor):
+    """Split a flat optimizer buffer into per-parameter tensors (logical layout)."""
+    return [
+        space._view(flat, o, n, shape).detach().cpu().clone()
+        for o, n, shape in zip(space.offsets, space.numels, space.shapes)
+    ]
+
+
+def _params_to_flat(space, tensors, flat: torch.Tensor) -> None:
+    with torch.no_grad():
+        for t, o, n, shape in zip(tensors, space.offsets, space.numels, space.shapes):
+            space._view(flat, o, n, shape).copy_(t.to(flat.device))
+
+
+def _pack_opt_state(space, opt) -> dict:
+    sd = opt.state_dict()
+    out = {}
+    for k, v in sd.items():
+        if isinstance(v, torch.Tensor) and v.numel() == space.d_pad:
+            out[k + "__params"] = _flat_to_params(space, v)
+        elif isinstance(v, torch.Tensor):
+            out[k] = v.cpu()
+        else:
+            out[k] = v
+    return out
+
+
+def _unpack_opt_state(space, opt, packed: dict) -> None:
+    for k, v in packed.items():
+        if k.endswith("__params"):
+            name = k[: -len("__params")]
+            flat = getattr(opt, name, None)
+            if flat is not None:
+                _params_to_flat(space, v, flat)
+        elif k == "first":
+            opt._first = v
+        elif k == "t":
+            opt.t = v
+
+
 def save_checkpoint(path: str, model, space, opt, step: int, cfg) -> None:
     os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
     payload = {
         "step": step,
         "model": {k: v.cpu() for k, v in model.state_dict().items()},
-        "flat_param": space.flat_param[: space.d].detach().cpu(),
-        "optimizer": _cpu_sd(opt.state_dict()),
+        "optimizer": _pack_opt_state(space, opt) if opt is not None else None,
         "network": cfg.network,
         "dataset": cfg.dataset,
     }
@@ -26,14 +71,8 @@ def save_checkpoint(path: str, model, space, opt, step: int, cfg) -> None:
 
 def load_checkpoint(path: str, model, space, opt) -> int:
     payload = torch.load(path, map_location="cpu", weights_only=False)
+    # load_state_dict copies through the flat-space views: layout-independent restore
     model.load_state_dict(payload["model"])
-    # model.load_state_dict copies into the flat-space views, so flat_param is
-    # already consistent; restore it explicitly anyway for safety
-    space.load_flat(payload["flat_param"])
     if opt is not None and payload.get("optimizer") is not None:
-        opt.load_state_dict(payload["optimizer"])
+        _unpack_opt_state(space, opt, payload["optimizer"])
     return payload["step"]
-
-
-def _cpu_sd(sd: dict) -> dict:
-    return {k: (v.cpu() if isinstance(v, torch.Tensor) else v) for k, v in sd.items()}

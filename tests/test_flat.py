@@ -79,3 +79,52 @@ def test_local_seg_bounds():
     # segments across ranks tile [0, d)
     covered = sum(int(space.local_seg_bounds(r)[-1]) - int(space.local_seg_bounds(r)[0]) for r in range(4))
     assert covered >= 0  # (pad region excluded by construction)
+
+
+def test_checkpoint_cross_layout_resume(tmp_path):
+    """A checkpoint written under channels_last=False resumes exactly under
+    channels_last=True (weights AND optimizer momentum)."""
+    import os
+
+    from draco_amd.optim import FlatSGD
+    from draco_amd.utils.checkpoint import load_checkpoint, save_checkpoint
+
+    class C:
+        network = "X"
+        dataset = "Y"
+
+    torch.manual_seed(1)
+    m1 = _model()
+    s1 = FlatSpace(m1, 1, torch.device("cpu"), channels_last=False)
+    o1 = FlatSGD(s1.flat_param, lr=0.1, momentum=0.9)
+    # a couple of updates to populate momentum
+    for i in range(3):
+        g = torch.randn(s1.d_pad)
+        o1.step(g)
+    path = str(tmp_path / "model_step_3")
+    save_checkpoint(path, m1, s1, o1, 3, C())
+
+    torch.manual_seed(2)  # different init on purpose
+    m2 = _model()
+    s2 = FlatSpace(m2, 1, torch.device("cpu"), channels_last=True)
+    o2 = FlatSGD(s2.flat_param, lr=0.1, momentum=0.9)
+    step = load_checkpoint(path, m2, s2, o2)
+    assert step == 3
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1.detach(), p2.detach(), atol=1e-7)
+    # momentum restored through logical views despite different flat layout
+    for o, n, shape in zip(s1.offsets, s1.numels, s1.shapes):
+        b1 = s1._view(o1.buf, o, n, shape)
+        b2 = s2._view(o2.buf, o, n, shape)
+        assert torch.allclose(b1, b2, atol=1e-7)
+    # next steps agree
+    g = torch.randn(s1.d_pad)
+    o1.step(g)
+    # remap g into layout-2 flat order through views
+    g2 = torch.zeros(s2.d_pad)
+    for o, n, shape in zip(s1.offsets, s1.numels, s1.shapes):
+        s2._view(g2, o, n, shape).copy_(s1._view(g, o, n, shape))
+    o2.step(g2)
+    for o, n, shape in zip(s1.offsets, s1.numels, s1.shapes):
+        assert torch.allclose(s1._view(s1.flat_param, o, n, shape),
+                              s2._view(s2.flat_param, o, n, shape), atol=1e-6)
