@@ -103,10 +103,3 @@ class FlatSpace:
     def shard_of(self, flat: torch.Tensor, rank: int) -> torch.Tensor:
         return flat[rank * self.shard : (rank + 1) * self.shard]
 
-    # ------------------------------------------------------------------ io
-    def state_dict_tensors(self):
-        return {"flat_param": self.flat_param[: self.d].clone()}
-
-    def load_flat(self, flat: torch.Tensor):
-        with torch.no_grad():
-            self.flat_param[: self.d].copy_(flat.to(self.device))

@@ -206,6 +206,10 @@ class Master(_PSBase):
         and treat them as ERASURES (the cyclic decode tolerates them as known error
         locations; a missing vote member simply loses its group's vote).
 
+        NOTE: the timeout path relies on blocking Work.wait() observed via waiter
+        threads, which is host-blocking on gloo (the CPU logic lane); NCCL p2p waits
+        are stream-ordered, so straggler_timeout is a gloo-lane feature.
+
         Protocol care: a timed-out worker's send still completes later (matched by the
         orphaned irecv).  The request is kept in self._stale[w]; at the next step the
         stale payload is DISCARDED and a fresh irecv posted, so the per-step message
