@@ -282,3 +282,30 @@ def test_adam_distributed():
     res = run_dist(_adam_worker, 2)
     assert res[0][2] == res[1][2], "params diverged across ranks under Adam"
     assert res[0][1] < res[0][0]
+
+
+def _cyclic_compress_worker(rank, world):
+    """bf16 wire compression on the cyclic (complex-plane) path: decode close to fp32."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    out = {}
+    for comp in ("none", "bf16"):
+        cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                     approach="cyclic", mode="cyclic", worker_fail=1, workers_per_rank=2,
+                     compress_grad=comp, max_steps=50, eval_freq=0, log_dir="",
+                     train_dir="/tmp/draco_ccomp")
+        t = Trainer(cfg)
+        t.logger.stdout_every = 0
+        for _ in range(4):
+            t.train_step()
+        out[comp] = t.space.flat_param.clone()
+        t.close()
+    diff = (out["none"] - out["bf16"]).abs().max()
+    scale = out["none"].abs().max()
+    assert diff < 0.05 * scale, float(diff)
+    return True
+
+
+def test_cyclic_bf16_wire_compression():
+    run_dist(_cyclic_compress_worker, 2)

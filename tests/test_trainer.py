@@ -152,3 +152,28 @@ def test_evaluator_polling_once(tmp_path):
     assert out.returncode == 0, out.stderr[-1500:]
     assert "Testset Performance: Cur Step:3" in out.stdout
     assert "Testset Performance: Cur Step:6" in out.stdout
+
+
+def test_nan_guard_skips_update(tmp_path):
+    """Failure detection: a non-finite decoded gradient must not touch the weights."""
+    cfg = _cfg(tmp_path, approach="baseline", mode="normal", worker_fail=0)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    t.train_step()
+    before = t.space.flat_param.clone()
+    orig = t.agg.aggregate
+
+    def poisoned(payload, step):
+        g = orig(payload, step)
+        g[5] = float("nan")
+        return g
+
+    t.agg.aggregate = poisoned
+    rec = t.train_step()
+    assert rec.get("skipped_update") is True
+    assert t.skipped_updates == 1
+    assert torch.equal(t.space.flat_param, before)
+    t.agg.aggregate = orig
+    rec = t.train_step()
+    assert "skipped_update" not in rec
+    t.close()
