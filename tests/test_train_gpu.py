@@ -45,6 +45,21 @@ def test_gpu_train_step(tmp_path, approach, mode, kw):
     t.close()
 
 
+def test_gpu_fp32_train(tmp_path):
+    """fp32 compute path (no autocast) with graphs: trains and vote holds (fp32
+    replica noise is ~1e-6 of row max; auto rtol 1e-4)."""
+    from draco_amd.parallel.trainer import Trainer
+
+    t = Trainer(_cfg(tmp_path, approach="maj_vote", mode="maj_vote", group_size=3,
+                     worker_fail=1, lr=0.02, dtype="fp32"))
+    t.logger.stdout_every = 0
+    losses = [t.train_step()["loss"] for _ in range(8)]
+    assert np.isfinite(losses).all()
+    assert min(losses[-3:]) < losses[0] * 2.0
+    assert t.skipped_updates == 0
+    t.close()
+
+
 def test_vote_tolerance_margin(tmp_path):
     """Honest replicas of the same batch must agree within the tolerance-vote
     threshold with wide margin, while a rev_grad adversary must be far outside it
