@@ -231,7 +231,7 @@ class Trainer:
     # replicas would run different kernels, their gradients would diverge beyond the
     # vote tolerance, and the majority vote breaks (observed: adversary winning 1/3
     # of votes).  Identical code across replicas is what the vote relies on.
-    def _run_fwd_bwd(self, key, grad_row: torch.Tensor, x, y, stream=None, freeze_bn=None):
+    def _run_fwd_bwd(self, key, grad_row: torch.Tensor, x, y, stream=None):
         if not self.use_graphs:
             return self._forward_backward(x, y, grad_row)
         g = self._graphs.get(key)
@@ -274,6 +274,10 @@ class Trainer:
     def _capture(self, grad_row: torch.Tensor, x, y):
         self.space.attach_grads(grad_row)
         static_x = x.clone()
+        if self.use_cl and static_x.dim() == 4:
+            # capture with channels_last inputs so MIOpen sees NHWC end-to-end
+            # (per-replay copy_ converts the incoming contiguous batch)
+            static_x = static_x.contiguous(memory_format=torch.channels_last)
         static_y = y.clone()
 
         def body():
