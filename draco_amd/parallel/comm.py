@@ -35,7 +35,9 @@ class Communicator:
                 backend=backend,
                 rank=rank,
                 world_size=world,
-                timeout=datetime.timedelta(seconds=300),
+                # generous watchdog: first-step skew across ranks includes MIOpen
+                # find + dynamo compile + graph capture (minutes on big models)
+                timeout=datetime.timedelta(seconds=1800),
             )
         self.backend = dist.get_backend() if self.distributed else "local"
 
