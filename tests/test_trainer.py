@@ -177,3 +177,16 @@ def test_nan_guard_skips_update(tmp_path):
     rec = t.train_step()
     assert "skipped_update" not in rec
     t.close()
+
+
+def test_repetition_r5_s2(tmp_path):
+    """r=5 group vote with two adversaries per step (config-4 shape) on CPU."""
+    t = Trainer(_cfg(tmp_path, approach="maj_vote", mode="maj_vote", group_size=5,
+                     worker_fail=2, err_mode="rev_grad"))
+    t.logger.stdout_every = 0
+    first = t.train_step()["loss"]
+    for _ in range(12):
+        last = t.train_step()["loss"]
+    assert last < first and np.isfinite(last)
+    assert t.agg.degenerate_steps == 0
+    t.close()
