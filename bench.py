@@ -55,7 +55,11 @@ def main():
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--channels-last", type=lambda v: v.lower() in ("1","true"), default=True)
     p.add_argument("--hip-graphs", type=lambda v: v.lower() in ("1","true"), default=True)
-    p.add_argument("--compile", type=lambda v: v.lower() in ("1","true"), default=True)
+    # OFF by default: torch.compile's GPU codegen (inductor) is Triton-backed, and
+    # this build's ground rule is hand-written HIP/CDNA4 + PyTorch-ROCm kernels only
+    # (BASELINE.json north star: "no Triton").  The flag documents the measured +27%
+    # for anyone who wants it.
+    p.add_argument("--compile", type=lambda v: v.lower() in ("1","true"), default=False)
     args = p.parse_args()
 
     from draco_amd.config import Config
