@@ -171,7 +171,13 @@ def main():
                 "images_processed_per_step": processed,
                 "seq_len": None,
                 "input": "3x32x32" if args.dataset == "Cifar10" else args.dataset,
-                "parallelism": f"coded-dp{world}(repetition r={args.group_size}, s={args.worker_fail}, {args.err_mode})",
+                "parallelism": (
+                    f"coded-dp{world}("
+                    + (f"repetition r={args.group_size}" if args.approach == "maj_vote"
+                       else f"cyclic r={2 * args.worker_fail + 1}" if args.approach == "cyclic"
+                       else f"baseline {args.mode}")
+                    + f", s={args.worker_fail}, {args.err_mode})"
+                ),
                 "topology": "colocated",
                 "optimizer": "fused SGD momentum=0.5",
             },
