@@ -181,7 +181,7 @@ def combine_rows(x: torch.Tensor, rows: torch.Tensor, w: torch.Tensor, out: torc
     if x.shape[1] == 0:
         out.zero_()
         return
-    torch.matmul(w, x[rows], out=out)
+    out.copy_((w @ x[rows]).reshape(-1))
 
 
 @torch.no_grad()
