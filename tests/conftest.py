@@ -2,8 +2,6 @@ import os
 import socket
 import sys
 
-import pytest
-
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
@@ -18,10 +16,3 @@ def free_port() -> int:
     s.close()
     return p
 
-
-@pytest.fixture
-def gloo_env(monkeypatch):
-    """Env template for multi-process gloo tests."""
-    monkeypatch.setenv("MASTER_ADDR", "127.0.0.1")
-    monkeypatch.setenv("MASTER_PORT", str(free_port()))
-    return dict(os.environ)
