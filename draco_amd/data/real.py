@@ -90,7 +90,7 @@ class RealClassification:
     rep_worker.py:89)."""
 
     def __init__(self, dataset: str, root: str, device: torch.device, train: bool = True,
-                 seed: int = 428, dtype=torch.float32):
+                 seed: int = 428, dtype=torch.float32, augment: bool = False):
         if dataset == "MNIST":
             x, y = load_mnist_idx(root, train)
         elif dataset == "Cifar10":
@@ -101,6 +101,11 @@ class RealClassification:
         self.y = torch.from_numpy(np.ascontiguousarray(y)).to(device)
         self.n = len(self.y)
         self.seed = seed
+        # train-time augmentation (reference util.py:29-65: RandomCrop(32, pad=4) +
+        # RandomHorizontalFlip for CIFAR).  Applied only to 3-channel 32x32 inputs;
+        # deterministic per batch start so repetition-group members on different
+        # ranks still draw bit-identical (augmented) batches.
+        self.augment = augment and train and dataset == "Cifar10"
         self._perm_epoch = -1
         self._perm = None
 
@@ -124,7 +129,26 @@ class RealClassification:
                 sel[m] = self._perm_for(e)[pos[m]]
         else:
             sel = self._perm_for(epoch)[pos]
-        return self.x[sel], self.y[sel]
+        x, y = self.x[sel], self.y[sel]
+        if self.augment:
+            x = self._augment_batch(x, start)
+        return x, y
+
+    def _augment_batch(self, x: torch.Tensor, start: int) -> torch.Tensor:
+        """RandomCrop(32, padding=4) + RandomHorizontalFlip, on device, seeded by the
+        batch start index — a pure function of (data, start) like get_batch itself."""
+        B, C, H, W = x.shape
+        g = torch.Generator(device=x.device)
+        g.manual_seed((self.seed * 0x9E3779B1 + start * 2654435761) & ((1 << 63) - 1))
+        xp = torch.nn.functional.pad(x, (4, 4, 4, 4))
+        offs = torch.randint(0, 9, (B, 2), generator=g, device=x.device)
+        flip = torch.rand(B, generator=g, device=x.device) < 0.5
+        ar = torch.arange(H, device=x.device)
+        idx_r = (offs[:, 0].view(B, 1, 1, 1) + ar.view(1, 1, H, 1)).expand(B, C, H, W + 8)
+        xr = xp.gather(2, idx_r)
+        idx_c = (offs[:, 1].view(B, 1, 1, 1) + ar.view(1, 1, 1, W)).expand(B, C, H, W)
+        xc = xr.gather(3, idx_c)
+        return torch.where(flip.view(B, 1, 1, 1), xc.flip(3), xc)
 
 
 class PrefetchLoader:

@@ -34,6 +34,17 @@ class SyntheticClassification:
         # fixed class means, modest separation so training has to work for it
         self.means = (torch.randn(classes, c, h, w, generator=g) * 0.7).to(device=device, dtype=dtype)
 
+    @staticmethod
+    def _mix(start: int) -> int:
+        """splitmix64 of the full 64-bit start index: distinct starts map to
+        distinct seeds with no structured collisions (the previous xor-fold let the
+        2**40 held-out eval offset collide with small training starts)."""
+        m = (1 << 64) - 1
+        z = (start + 0x9E3779B97F4A7C15) & m
+        z = ((z ^ (z >> 30)) * 0xBF58476D1CE4E5B9) & m
+        z = ((z ^ (z >> 27)) * 0x94D049BB133111EB) & m
+        return (z ^ (z >> 31)) & ((1 << 63) - 1)  # manual_seed wants non-negative int64
+
     def get_batch(self, start: int, batch: int):
         """Deterministic batch for global sample indices [start, start+batch).
 
@@ -41,7 +52,7 @@ class SyntheticClassification:
         group members on different GPUs draw bit-identical batches with no host
         round-trip — generation costs <1 ms vs ~25 ms for host randn + H2D copy.
         """
-        seed = 0x9E3779B9 ^ (start & 0xFFFFFFFF) ^ ((start >> 32) << 1)
+        seed = self._mix(start)
         g = torch.Generator(device=self.device)
         g.manual_seed(seed)
         y = torch.randint(0, self.classes, (batch,), generator=g, device=self.device)

@@ -91,21 +91,47 @@ class Aggregator:
 
 
 class MeanAggregator(Aggregator):
-    """Plain averaging over all P = L*world logical workers (reduce-scatter path:
-    no all_to_all needed since mean commutes with sharding)."""
+    """Plain averaging over all P = L*world logical workers.
+
+    Mean commutes with sharding, so no all_to_all is needed: start_row posts an
+    async per-row reduce_scatter (the sum happens in-flight on the wire), which
+    both gives the baseline path the same backward/comm overlap as the coded
+    paths and ships each byte exactly once (round-1 bug: the base-class
+    all_to_all was posted but never awaited — dead traffic + unbounded _works)."""
 
     name = "mean"
 
     def __init__(self, comm, space, num_workers: int):
         super().__init__(comm, space)
         self.num_workers = num_workers
+        self._rs_shards = None  # (rows, shard) per-row reduce_scatter outputs
+
+    def start_row(self, payload: torch.Tensor, row: int) -> None:
+        if not self.comm.distributed:
+            return
+        if self._rs_shards is None or self._rs_shards.shape[0] != payload.shape[0]:
+            self._rs_shards = torch.empty(payload.shape[0], self.space.shard,
+                                          dtype=torch.float32, device=payload.device)
+        w = self.comm.reduce_scatter_row(payload[row], self._rs_shards[row], async_op=True)
+        if w is not None:
+            self._works.append(w)
+        self._started.add(row)
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
-        local_sum = self._out  # reuse as scratch for the local sum
-        ops.sum_rows(payload, local_sum)
-        shard_sum = self.comm.reduce_scatter_sum(local_sum)
-        shard_sum /= float(self.num_workers)
-        self.comm.all_gather_shard(shard_sum, self._out)
+        if not self.comm.distributed:
+            ops.sum_rows(payload, self._out)
+            self._out /= float(self.num_workers)
+            return self._out
+        for r in range(payload.shape[0]):
+            if r not in self._started:
+                self.start_row(payload, r)
+        for w in self._works:
+            w.wait()
+        self._works = []
+        self._started = set()
+        torch.sum(self._rs_shards, dim=0, out=self._shard_out)
+        self._shard_out /= float(self.num_workers)
+        self.comm.all_gather_shard(self._shard_out, self._out)
         return self._out
 
 

@@ -85,6 +85,17 @@ class Communicator:
                 w.wait()
         return recv.view(L * self.world, shard)
 
+    def reduce_scatter_row(self, payload_row: torch.Tensor, out_shard: torch.Tensor,
+                           async_op: bool = False):
+        """payload_row (d_pad,) -> out_shard (shard,) = global sum of this rank's
+        shard of the row.  The overlap-capable form of reduce_scatter_sum: the
+        trainer posts one per payload row as its backward completes."""
+        if not self.distributed:
+            out_shard.copy_(payload_row)
+            return None
+        work = dist.reduce_scatter_tensor(out_shard, payload_row, async_op=async_op)
+        return work if async_op else None
+
     def reduce_scatter_sum(self, payload_sum: torch.Tensor) -> torch.Tensor:
         """payload_sum: (d_pad,) local sum -> (shard,) global sum of this rank's shard."""
         shard = payload_sum.shape[0] // self.world

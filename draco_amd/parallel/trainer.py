@@ -106,12 +106,18 @@ class Trainer:
             rtol = cfg.vote_rtol
             if rtol < 0:  # auto: bitwise on CPU, tolerance on GPU (see VoteAggregator doc)
                 if device.type != "cuda":
-                    rtol = 0.0
+                    rtol = 0.0  # CPU autograd is reproducible; identical fp32 rows
+                    # cast to identical bf16 rows, so a bf16 wire stays bitwise too
                 elif cfg.dtype == "bf16":
                     # bf16 autocast: replica noise measured at 1.2-3.5% of the row max
                     # under MIOpen algo/order variation (tools/diag_det.py); 0.2 keeps
                     # >5x margin while a rev_grad adversary is ~500x outside
                     rtol = 2e-1
+                elif self._comm_dtype() == torch.bfloat16:
+                    # fp32 compute but bf16 wire: GPU fp-reorder noise can push
+                    # honest replicas across bf16 rounding boundaries (1 ulp =
+                    # 2^-8 relative), far above the fp32 reorder level
+                    rtol = 1e-2
                 else:
                     rtol = 1e-4  # fp32 reorder noise
             self.vote_rtol = rtol
@@ -183,11 +189,14 @@ class Trainer:
         # "compress" = reference CLI alias (README.md:118, blosc) -> bf16 wire dtype
         return torch.bfloat16 if self.cfg.compress_grad in ("bf16", "compress") else torch.float32
 
-    def _dataset(self):
+    def _dataset(self, train: bool = True):
         from ..data.real import RealClassification, dataset_available
 
         if dataset_available(self.cfg.dataset, self.cfg.data_root):
-            return RealClassification(self.cfg.dataset, self.cfg.data_root, self.device)
+            return RealClassification(self.cfg.dataset, self.cfg.data_root, self.device,
+                                      train=train, augment=train)
+        # held-out synthetic stream: same class means (the task), disjoint index
+        # space (the 2**40 offset in evaluate(); splitmix64 seeding cannot collide)
         return SyntheticClassification(self.cfg.dataset, self.device, seed=1234)
 
     def _baseline_aggregator(self, cfg: Config):
@@ -398,8 +407,10 @@ class Trainer:
             rec["gpu_agg"] = ev_comp.elapsed_time(ev_agg) / 1e3
             rec["gpu_update"] = ev_agg.elapsed_time(ev_end) / 1e3
         if cfg.eval_freq > 0 and self.step_num % cfg.eval_freq == 0:
-            if self.rank == 0:
-                self.save()
+            # collective on ALL ranks: save() runs sync_buffers (an all_reduce);
+            # only the file write inside is rank-0-gated.  A rank-0-only call here
+            # deadlocks world>1 (the other ranks enter the next step's collectives).
+            self.save()
         self.logger.log(rec)
         return rec
 
@@ -423,7 +434,7 @@ class Trainer:
         """Held-out Prec@1/Prec@5 + loss (reference eval cadence,
         baseline_worker.py:148-155 / distributed_evaluator.py:92-110)."""
         self.model.eval()
-        data = self._dataset()
+        data = self._dataset(train=False)  # test split for real data (held out)
         correct = correct5 = total = 0
         loss_sum = 0.0
         for b in range(n_batches):
@@ -462,8 +473,9 @@ class Trainer:
                 off += n
 
     def save(self):
-        self.sync_buffers()
-        save_checkpoint(self._ckpt_path(), self.model, self.space, self.opt, self.step_num, self.cfg)
+        self.sync_buffers()  # collective — every rank must call save() together
+        if self.rank == 0:
+            save_checkpoint(self._ckpt_path(), self.model, self.space, self.opt, self.step_num, self.cfg)
 
     def load(self, step: int):
         load_checkpoint(self._ckpt_path(step), self.model, self.space, self.opt)

@@ -55,11 +55,22 @@ def _unpack_opt_state(space, opt, packed: dict) -> None:
             opt.t = v
 
 
+_COMPILE_PREFIX = "_orig_mod."
+
+
+def _strip_compile_prefix(sd: dict) -> dict:
+    """torch.compile wraps the module, prefixing state_dict keys with `_orig_mod.`;
+    checkpoints are stored with canonical keys so they are interchangeable across
+    compile settings."""
+    return {k[len(_COMPILE_PREFIX):] if k.startswith(_COMPILE_PREFIX) else k: v
+            for k, v in sd.items()}
+
+
 def save_checkpoint(path: str, model, space, opt, step: int, cfg) -> None:
     os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
     payload = {
         "step": step,
-        "model": {k: v.cpu() for k, v in model.state_dict().items()},
+        "model": {k: v.cpu() for k, v in _strip_compile_prefix(model.state_dict()).items()},
         "optimizer": _pack_opt_state(space, opt) if opt is not None else None,
         "network": cfg.network,
         "dataset": cfg.dataset,
@@ -71,8 +82,11 @@ def save_checkpoint(path: str, model, space, opt, step: int, cfg) -> None:
 
 def load_checkpoint(path: str, model, space, opt) -> int:
     payload = torch.load(path, map_location="cpu", weights_only=False)
+    sd = _strip_compile_prefix(payload["model"])  # tolerate old compiled checkpoints
+    if any(k.startswith(_COMPILE_PREFIX) for k in model.state_dict()):
+        sd = {_COMPILE_PREFIX + k: v for k, v in sd.items()}  # loading INTO a compiled model
     # load_state_dict copies through the flat-space views: layout-independent restore
-    model.load_state_dict(payload["model"])
+    model.load_state_dict(sd)
     if opt is not None and payload.get("optimizer") is not None:
         _unpack_opt_state(space, opt, payload["optimizer"])
     return payload["step"]
