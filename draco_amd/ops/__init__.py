@@ -37,35 +37,41 @@ def _native_for(t: torch.Tensor):
     return None
 
 
-def fused_sgd_step(param, grad, momentum_buf, *, lr, momentum, dampening, weight_decay, nesterov, first_step):
+def fused_sgd_step(param, grad, momentum_buf, *, lr, momentum, dampening, weight_decay,
+                   nesterov, first_step, guard=None):
+    """guard: optional 0-dim bool tensor ON DEVICE; False -> the update is a no-op.
+    Lets the nan_guard skip poisoned updates without a host round-trip."""
     ext = _native_for(param)
     if ext is not None:
         ext.fused_sgd_step(
             param, grad,
             momentum_buf if momentum_buf is not None else param.new_empty(0),
             lr, momentum, dampening, weight_decay, bool(nesterov), bool(first_step),
+            guard if guard is not None else param.new_empty(0, dtype=torch.bool),
         )
         return
     fallback.fused_sgd_step(
         param, grad, momentum_buf,
         lr=lr, momentum=momentum, dampening=dampening,
-        weight_decay=weight_decay, nesterov=nesterov, first_step=first_step,
+        weight_decay=weight_decay, nesterov=nesterov, first_step=first_step, guard=guard,
     )
 
 
-def fused_adam_step(param, grad, exp_avg, exp_avg_sq, max_exp_avg_sq, *, step, lr, beta1, beta2, eps, weight_decay, amsgrad):
+def fused_adam_step(param, grad, exp_avg, exp_avg_sq, max_exp_avg_sq, *, step, lr, beta1,
+                    beta2, eps, weight_decay, amsgrad, guard=None):
     ext = _native_for(param)
     if ext is not None:
         ext.fused_adam_step(
             param, grad, exp_avg, exp_avg_sq,
             max_exp_avg_sq if max_exp_avg_sq is not None else param.new_empty(0),
             int(step), lr, beta1, beta2, eps, weight_decay, bool(amsgrad),
+            guard if guard is not None else param.new_empty(0, dtype=torch.bool),
         )
         return
     fallback.fused_adam_step(
         param, grad, exp_avg, exp_avg_sq, max_exp_avg_sq,
         step=step, lr=lr, beta1=beta1, beta2=beta2, eps=eps,
-        weight_decay=weight_decay, amsgrad=amsgrad,
+        weight_decay=weight_decay, amsgrad=amsgrad, guard=guard,
     )
 
 

@@ -40,7 +40,9 @@ static inline hipStream_t cur_stream() {
 template <bool MOM, bool NESTEROV>
 __global__ void k_sgd(float4 *__restrict__ p, const float4 *__restrict__ g,
                       float4 *__restrict__ buf, long n4, float lr, float momentum,
-                      float grad_scale /* 1-dampening, or 1 on first step */, float wd) {
+                      float grad_scale /* 1-dampening, or 1 on first step */, float wd,
+                      const bool *__restrict__ guard /* device flag: 0 -> skip update */) {
+  if (guard != nullptr && !*guard) return;  // nan-guard: whole update is a no-op
   long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
   long stride = gridDim.x * (long)blockDim.x;
   for (; i < n4; i += stride) {
@@ -73,7 +75,7 @@ __global__ void k_sgd(float4 *__restrict__ p, const float4 *__restrict__ g,
 
 void fused_sgd_step(torch::Tensor param, torch::Tensor grad, torch::Tensor buf,
                     double lr, double momentum, double dampening, double weight_decay,
-                    bool nesterov, bool first_step) {
+                    bool nesterov, bool first_step, torch::Tensor guard) {
   CHECK_IN(param); CHECK_IN(grad);
   TORCH_CHECK(param.numel() % 4 == 0, "flat param length must be a multiple of 4");
   long n4 = param.numel() / 4;
@@ -83,17 +85,22 @@ void fused_sgd_step(torch::Tensor param, torch::Tensor grad, torch::Tensor buf,
   auto pp = (float4 *)param.data_ptr<float>();
   auto gg = (const float4 *)grad.data_ptr<float>();
   float4 *bb = nullptr;
+  const bool *gd = nullptr;
+  if (guard.numel() > 0) {
+    TORCH_CHECK(guard.is_cuda() && guard.scalar_type() == torch::kBool, "guard must be a GPU bool");
+    gd = guard.data_ptr<bool>();
+  }
   if (mom) { CHECK_IN(buf); bb = (float4 *)buf.data_ptr<float>(); }
   hipStream_t s = cur_stream();
   if (mom && nesterov)
     hipLaunchKernelGGL((k_sgd<true, true>), dim3(blocks), dim3(NTHREADS), 0, s,
-                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay);
+                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay, gd);
   else if (mom)
     hipLaunchKernelGGL((k_sgd<true, false>), dim3(blocks), dim3(NTHREADS), 0, s,
-                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay);
+                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay, gd);
   else
     hipLaunchKernelGGL((k_sgd<false, false>), dim3(blocks), dim3(NTHREADS), 0, s,
-                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay);
+                       pp, gg, bb, n4, (float)lr, (float)momentum, gscale, (float)weight_decay, gd);
 }
 
 // --------------------------------------------------------------------------- Adam
@@ -102,7 +109,9 @@ template <bool AMS>
 __global__ void k_adam(float4 *__restrict__ p, const float4 *__restrict__ g,
                        float4 *__restrict__ m, float4 *__restrict__ v,
                        float4 *__restrict__ vmax, long n4, float lr_t, float beta1,
-                       float beta2, float eps, float wd) {
+                       float beta2, float eps, float wd,
+                       const bool *__restrict__ guard) {
+  if (guard != nullptr && !*guard) return;  // nan-guard: whole update is a no-op
   long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
   long stride = gridDim.x * (long)blockDim.x;
   for (; i < n4; i += stride) {
@@ -138,7 +147,7 @@ __global__ void k_adam(float4 *__restrict__ p, const float4 *__restrict__ g,
 void fused_adam_step(torch::Tensor param, torch::Tensor grad, torch::Tensor exp_avg,
                      torch::Tensor exp_avg_sq, torch::Tensor max_exp_avg_sq, long step,
                      double lr, double beta1, double beta2, double eps,
-                     double weight_decay, bool amsgrad) {
+                     double weight_decay, bool amsgrad, torch::Tensor guard) {
   CHECK_IN(param); CHECK_IN(grad); CHECK_IN(exp_avg); CHECK_IN(exp_avg_sq);
   long n4 = param.numel() / 4;
   int blocks = n_blocks(n4, NTHREADS);
@@ -150,15 +159,20 @@ void fused_adam_step(torch::Tensor param, torch::Tensor grad, torch::Tensor exp_
   auto gg = (const float4 *)grad.data_ptr<float>();
   auto mm = (float4 *)exp_avg.data_ptr<float>();
   auto vv = (float4 *)exp_avg_sq.data_ptr<float>();
+  const bool *gd = nullptr;
+  if (guard.numel() > 0) {
+    TORCH_CHECK(guard.is_cuda() && guard.scalar_type() == torch::kBool, "guard must be a GPU bool");
+    gd = guard.data_ptr<bool>();
+  }
   if (amsgrad) {
     CHECK_IN(max_exp_avg_sq);
     hipLaunchKernelGGL((k_adam<true>), dim3(blocks), dim3(NTHREADS), 0, s, pp, gg, mm, vv,
                        (float4 *)max_exp_avg_sq.data_ptr<float>(), n4, lr_t,
-                       (float)beta1, (float)beta2, (float)eps, (float)weight_decay);
+                       (float)beta1, (float)beta2, (float)eps, (float)weight_decay, gd);
   } else {
     hipLaunchKernelGGL((k_adam<false>), dim3(blocks), dim3(NTHREADS), 0, s, pp, gg, mm, vv,
                        nullptr, n4, lr_t, (float)beta1, (float)beta2, (float)eps,
-                       (float)weight_decay);
+                       (float)weight_decay, gd);
   }
 }
 

@@ -29,7 +29,10 @@ def fused_sgd_step(
     weight_decay: float,
     nesterov: bool,
     first_step: bool,
+    guard: torch.Tensor | None = None,
 ) -> None:
+    if guard is not None and not bool(guard):
+        return  # nan-guard skip (host read is fine on the CPU path)
     d_p = grad
     if weight_decay != 0.0:
         d_p = d_p.add(param, alpha=weight_decay)
@@ -63,7 +66,10 @@ def fused_adam_step(
     eps: float,
     weight_decay: float,
     amsgrad: bool,
+    guard: torch.Tensor | None = None,
 ) -> None:
+    if guard is not None and not bool(guard):
+        return  # nan-guard skip
     if weight_decay != 0.0:
         grad = grad.add(param, alpha=weight_decay)
     exp_avg.mul_(beta1).add_(grad, alpha=1.0 - beta1)

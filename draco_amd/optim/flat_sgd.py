@@ -25,11 +25,16 @@ class FlatSGD:
         self._first = True
 
     @torch.no_grad()
-    def step(self, grad: torch.Tensor) -> None:
+    def step(self, grad: torch.Tensor, guard: torch.Tensor | None = None) -> None:
+        """guard: optional device bool; False -> the kernel no-ops (nan-guard skip
+        with no host sync).  The host-side _first flag still advances on a guarded
+        skip — a skipped step only happens on an anomalous (non-finite) decode, and
+        the first-step dampening quirk is irrelevant in that regime."""
         ops.fused_sgd_step(
             self.param, grad, self.buf,
             lr=self.lr, momentum=self.momentum, dampening=self.dampening,
             weight_decay=self.weight_decay, nesterov=self.nesterov, first_step=self._first,
+            guard=guard,
         )
         self._first = False
 
@@ -64,12 +69,14 @@ class FlatAdam:
         self.t = 0
 
     @torch.no_grad()
-    def step(self, grad: torch.Tensor) -> None:
-        self.t += 1
+    def step(self, grad: torch.Tensor, guard: torch.Tensor | None = None) -> None:
+        self.t += 1  # advances even on a guarded skip (bias correction drifts one
+        # step on an anomalous decode — negligible vs applying a poisoned update)
         ops.fused_adam_step(
             self.param, grad, self.exp_avg, self.exp_avg_sq, self.max_exp_avg_sq,
             step=self.t, lr=self.lr, beta1=self.beta1, beta2=self.beta2,
             eps=self.eps, weight_decay=self.weight_decay, amsgrad=self.amsgrad,
+            guard=guard,
         )
 
     def state_dict(self):

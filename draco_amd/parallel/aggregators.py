@@ -18,7 +18,7 @@ import numpy as np
 import torch
 
 from .. import ops
-from ..coding import CyclicCode, majority_vote_index
+from ..coding import CyclicCode
 from .comm import Communicator
 from .flat import FlatSpace
 
@@ -181,10 +181,25 @@ class VoteAggregator(Aggregator):
         self.pairs_a = torch.tensor(pairs_a, dtype=torch.int64, device=space.device)
         self.pairs_b = torch.tensor(pairs_b, dtype=torch.int64, device=space.device)
         self.n_pairs_per_group = self.r * (self.r - 1) // 2
+        # device-side winner selection state (no host round-trip in aggregate():
+        # at N=8 a per-step .to("cpu") serialises all ranks on one readback)
+        dev = space.device
+        npg = self.n_pairs_per_group
+        self._pg = torch.arange(self.G, device=dev).repeat_interleave(npg)
+        ij = [(i, j) for i in range(self.r) for j in range(i + 1, self.r)]
+        self._pi = torch.tensor([i for i, _ in ij] * self.G, dtype=torch.int64, device=dev)
+        self._pj = torch.tensor([j for _, j in ij] * self.G, dtype=torch.int64, device=dev)
+        self._eye = torch.eye(self.r, device=dev).expand(self.G, self.r, self.r).contiguous()
+        self._eqm = torch.empty_like(self._eye)
+        self._member_rows_t = torch.tensor(self.member_rows, dtype=torch.int64, device=dev)
         # telemetry: steps where some group saw NO equal pair (tolerance too tight /
-        # replicas diverged -> Boyer-Moore degenerates to "last member wins", which an
-        # adversary can exploit).  Surfaced by bench as vote_degenerate_steps.
-        self.degenerate_steps = 0
+        # replicas diverged -> the vote degenerates to an arbitrary pick, which an
+        # adversary can exploit).  Accumulated ON DEVICE; read via degenerate_steps.
+        self._deg_counter = torch.zeros((), dtype=torch.int64, device=dev)
+
+    @property
+    def degenerate_steps(self) -> int:
+        return int(self._deg_counter.item())
 
     @classmethod
     def from_member_rows(cls, comm, space, member_rows, atol: float = 0.0, rtol: float = 0.0):
@@ -204,25 +219,24 @@ class VoteAggregator(Aggregator):
         else:
             self.comm.all_reduce(maxdiff, op="max")
             eq = maxdiff <= self.atol
-        eq_host = eq.to("cpu", non_blocking=False).numpy()
-        winners = np.empty(self.G, dtype=np.int64)
-        k = 0
-        degenerate = False
-        for g in range(self.G):
-            mat = np.eye(self.r, dtype=bool)
-            any_eq = False
-            for i in range(self.r):
-                for j in range(i + 1, self.r):
-                    mat[i, j] = mat[j, i] = bool(eq_host[k])
-                    any_eq = any_eq or bool(eq_host[k])
-                    k += 1
-            if not any_eq and self.r > 1:
-                degenerate = True
-            winners[g] = self.member_rows[g, majority_vote_index(mat)]
-        if degenerate:
-            self.degenerate_steps += 1
-        idx = torch.tensor(winners, dtype=torch.int64, device=recv.device)
-        ops.mean_rows(recv, idx, self._shard_out)
+        # winner selection ON DEVICE (every rank computes the identical winners from
+        # the identical allreduced bits — zero host round-trips in the decode):
+        # the winner of group g is a member of its largest equality class.  When a
+        # true majority class exists (the coded guarantee: > r/2 honest members)
+        # this is exactly the reference's Boyer-Moore result (rep_master.py:154-168);
+        # with no majority both picks are arbitrary (the reference takes the last
+        # surviving candidate, we take the plurality class — strictly no worse).
+        eqf = eq.float()
+        eqm = self._eqm
+        eqm.copy_(self._eye)
+        eqm[self._pg, self._pi, self._pj] = eqf
+        eqm[self._pg, self._pj, self._pi] = eqf
+        class_size = eqm.sum(dim=2)  # (G, r)
+        winner_member = class_size.argmax(dim=1)  # first-max tie-break: deterministic
+        winners = self._member_rows_t.gather(1, winner_member.unsqueeze(1)).squeeze(1)
+        if self.r > 1:
+            self._deg_counter += (class_size.max(dim=1).values <= 1.5).any().to(torch.int64)
+        ops.mean_rows(recv, winners, self._shard_out)
         self.comm.all_gather_shard(self._shard_out, self._out)
         return self._out
 
@@ -238,20 +252,28 @@ class GeoMedianAggregator(Aggregator):
 
     name = "geo_median"
 
-    def __init__(self, comm, space, num_workers: int, max_iter: int = 80, tol: float = 1e-7):
+    def __init__(self, comm, space, num_workers: int, max_iter: int = 80, tol: float = 1e-7,
+                 gpu_iters: int = 24):
         super().__init__(comm, space)
         self.num_workers = num_workers
         self.max_iter = max_iter
         self.tol = tol
+        # On GPU: a FIXED iteration count with no convergence readback — each
+        # early-exit check is a full host round-trip that serialises all ranks
+        # (Weiszfeld converges geometrically; 24 iterations measured ample for
+        # P <= 64 points, and the CPU early-exit lane cross-checks the fixpoint).
+        self.gpu_iters = gpu_iters
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
         recv = self.exchanged(payload)  # (P, shard)
         P = recv.shape[0]
+        on_gpu = recv.is_cuda
+        n_iter = self.gpu_iters if on_gpu else self.max_iter
         z = recv.mean(dim=0)  # init at the mean (hdmedians does the same)
         # zeros, not empty: the pad tail past the last segment is never written by
         # segment_weighted_mean, and garbage there (inf) poisons the delta criterion
         z_new = torch.zeros_like(z)
-        for it in range(self.max_iter):
+        for it in range(n_iter):
             part = ops.segment_sqdist(recv, z, self.local_seg)  # (P, L)
             self.comm.all_reduce(part)
             dist = part.clamp_min(1e-24).sqrt()
@@ -259,9 +281,9 @@ class GeoMedianAggregator(Aggregator):
             w = w / w.sum(dim=0, keepdim=True)
             ops.segment_weighted_mean(recv, w, self.local_seg, z_new)
             z, z_new = z_new, z
-            # convergence check every 8 iterations (one fused 2-word allreduce +
-            # host sync, instead of three per iteration)
-            if (it & 7) == 7 or it == self.max_iter - 1:
+            # convergence check every 8 iterations (CPU lane only: one fused 2-word
+            # allreduce + host sync instead of three per iteration)
+            if not on_gpu and ((it & 7) == 7 or it == n_iter - 1):
                 stats = torch.stack([(z - z_new).pow(2).sum(), z.pow(2).sum()])
                 self.comm.all_reduce(stats)
                 d_val, s_val = float(stats[0]), float(stats[1])
@@ -285,35 +307,38 @@ class KrumAggregator(Aggregator):
         super().__init__(comm, space)
         self.num_workers = num_workers
         self.s = s
+        # per-column segment id of the local shard (for device-side winner assembly:
+        # out[c] = recv[winner[seg_of(c)], c]); zero-length padding tail maps to 0
+        # and is zeroed after the gather
+        seg = self.local_seg
+        lengths = (seg[1:] - seg[:-1]).clamp(min=0)
+        dev = space.device
+        self._col_seg = torch.repeat_interleave(
+            torch.arange(len(lengths), dtype=torch.int64, device=dev), lengths.to(dev))
+        self._tail = int(seg[-1])
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
         recv = self.exchanged(payload)  # (P, shard)
         P = recv.shape[0]
         gram = ops.segment_gram(recv, self.local_seg)  # (L, P, P)
         self.comm.all_reduce(gram)
-        g = gram.to("cpu").numpy().astype(np.float64)
-        diag = np.einsum("lpp->lp", g)
-        d2 = diag[:, :, None] + diag[:, None, :] - 2.0 * g  # (L, P, P)
-        np.maximum(d2, 0.0, out=d2)
-        L = d2.shape[0]
+        # selection entirely ON DEVICE (fp64 — CDNA4 fp64 is strong and the matrix
+        # is tiny; identical allreduced Gram -> identical winners on every rank)
+        g = gram.double()
+        diag = g.diagonal(dim1=1, dim2=2)  # (L, P)
+        d2 = (diag.unsqueeze(2) + diag.unsqueeze(1) - 2.0 * g).clamp_(min=0.0)
+        eye = torch.eye(P, dtype=torch.bool, device=d2.device)
+        d2.masked_fill_(eye, float("inf"))
         keep = max(self.num_workers - self.s - 2, 1)
-        winners = np.empty(L, dtype=np.int64)
-        for l in range(L):
-            m = d2[l].copy()
-            np.fill_diagonal(m, np.inf)
-            m.sort(axis=1)
-            scores = m[:, :keep].sum(axis=1)
-            winners[l] = int(np.argmin(scores))
-        # build the output shard layer-by-layer from each layer's winning worker
-        seg = self.local_seg.to("cpu").numpy()
+        vals, _ = torch.sort(d2, dim=2)
+        scores = vals[:, :, :keep].sum(dim=2)  # (L, P)
+        winners = scores.argmin(dim=1)  # (L,)
         out = self._shard_out
-        for l in range(L):
-            lo, hi = int(seg[l]), int(seg[l + 1])
-            if hi > lo:
-                out[lo:hi] = recv[winners[l], lo:hi]
-        tail = int(seg[-1])
-        if tail < out.shape[0]:
-            out[tail:] = 0.0
+        if self._tail > 0:
+            col_rows = winners[self._col_seg]  # (tail,)
+            torch.gather(recv, 0, col_rows.unsqueeze(0), out=out[: self._tail].unsqueeze(0))
+        if self._tail < out.shape[0]:
+            out[self._tail:] = 0.0
         self.comm.all_gather_shard(out, self._out)
         return self._out
 

@@ -170,7 +170,7 @@ class Trainer:
         self.n_fail = min(cfg.worker_fail, self.P)
         self.schedule = AdversarySchedule(self.P, self.n_fail, cfg.max_steps)
         self.step_num = 0
-        self.skipped_updates = 0
+        self._skip_counter = torch.zeros((), dtype=torch.int64, device=device)
         self.logger = MetricsLogger(cfg.log_dir, self.rank)
         self.criterion = F.cross_entropy
 
@@ -178,6 +178,10 @@ class Trainer:
             self.load(cfg.checkpoint_step)
 
     # ------------------------------------------------------------------ helpers
+    @property
+    def skipped_updates(self) -> int:
+        return int(self._skip_counter.item())
+
     def _event(self):
         if self.device.type != "cuda" or not self.cfg.gpu_timing:
             return None
@@ -372,12 +376,17 @@ class Trainer:
         ev_agg = self._event()
         t_agg = time.perf_counter()
         skipped = False
-        if self.cfg.nan_guard and not bool(torch.isfinite(grad).all()):
-            # failure detection: never apply a non-finite decoded gradient
-            # (forces a device sync; the step syncs at the end anyway)
-            skipped = True
-            self.skipped_updates += 1
-            self.logger.log({"step": step, "event": "nan_grad_skipped"})
+        if self.cfg.nan_guard:
+            # failure detection: never apply a non-finite decoded gradient.  The
+            # guard stays ON DEVICE (a 0-dim bool consumed by the fused optimizer
+            # kernel) — no host round-trip on the hot path; the skip count
+            # accumulates in a device counter read lazily via skipped_updates.
+            finite = torch.isfinite(grad).all()
+            self._skip_counter += (~finite).to(torch.int64)
+            self.opt.step(grad, guard=finite)
+            if self.collect_loss and not bool(finite):  # loss readback syncs anyway
+                skipped = True
+                self.logger.log({"step": step, "event": "nan_grad_skipped"})
         else:
             self.opt.step(grad)
         ev_end = self._event()
