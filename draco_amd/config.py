@@ -49,6 +49,9 @@ class Config:
     hip_graphs: bool = True       # capture fwd+bwd in hipGraphs (launch-bound models; GPU only)
     straggler_timeout: float = 0.0  # ps topology: seconds after first gradient before
                                     # missing workers become erasures (0 = wait forever)
+    bucket_mb: float = 25.0       # per-layer overlap: backward hooks ship the payload
+                                  # row in ~bucket_mb chunks while backward continues
+                                  # (eager path, baseline/maj_vote; 0 = whole-row)
     nan_guard: bool = True        # failure detection: skip updates on non-finite decode
     gpu_timing: bool = False      # device-accurate phase spans via HIP events (metrics)
     compile: bool = False         # torch.compile the model before hipGraph capture

@@ -46,19 +46,25 @@ class Aggregator:
         self._recv32 = None
         self._works: list = []
         self._started: set = set()
+        self._pending_copies: list = []  # (tmp2d, row, slo, shi) from bucket exchanges
 
     # ---------------------------------------------------------- row exchange
+    def _ensure_recv(self, rows: int, device) -> None:
+        if self._recv is None or self._recv.shape[0] != rows:
+            # zeros, not empty: the bucketed path never transmits the [d, d_pad)
+            # padding tail (payload padding is always zero), so recv must start zero
+            self._recv = torch.zeros(rows, self.comm.world, self.space.shard,
+                                     dtype=self.comm_dtype, device=device)
+            if self.comm_dtype != torch.float32:
+                self._send = torch.empty(rows, self.space.d_pad,
+                                         dtype=self.comm_dtype, device=device)
+                self._recv32 = torch.empty(rows, self.comm.world, self.space.shard,
+                                           dtype=torch.float32, device=device)
+
     def start_row(self, payload: torch.Tensor, row: int) -> None:
         if not self.comm.distributed:
             return
-        if self._recv is None or self._recv.shape[0] != payload.shape[0]:
-            self._recv = torch.empty(payload.shape[0], self.comm.world, self.space.shard,
-                                     dtype=self.comm_dtype, device=payload.device)
-            if self.comm_dtype != torch.float32:
-                self._send = torch.empty(payload.shape[0], payload.shape[1],
-                                         dtype=self.comm_dtype, device=payload.device)
-                self._recv32 = torch.empty(payload.shape[0], self.comm.world, self.space.shard,
-                                           dtype=torch.float32, device=payload.device)
+        self._ensure_recv(payload.shape[0], payload.device)
         if self.comm_dtype != torch.float32:
             self._send[row].copy_(payload[row])  # fp32 -> bf16 cast on device
             send_row = self._send[row]
@@ -67,6 +73,35 @@ class Aggregator:
         w = self.comm.all_to_all_row(send_row, self._recv[row], async_op=True)
         if w is not None:
             self._works.append(w)
+        self._started.add(row)
+
+    def start_bucket(self, payload: torch.Tensor, row: int, lo: int, hi: int) -> None:
+        """Exchange just the [lo, hi) flat range of one payload row (posted from a
+        post-accumulate-grad hook while the rest of backward still computes — the
+        per-layer comm/compute overlap, reference lenet.py:114-218 interleaved
+        isends).  Same wire format as start_row, delivered into the same recv slots;
+        the trainer calls mark_row_started(row) once every bucket of the row is
+        posted."""
+        if not self.comm.distributed:
+            return
+        self._ensure_recv(payload.shape[0], payload.device)
+        world, shard, rank = self.comm.world, self.space.shard, self.comm.rank
+        if self.comm_dtype != torch.float32:
+            self._send[row, lo:hi].copy_(payload[row, lo:hi])
+            src = self._send[row, lo:hi]
+        else:
+            src = payload[row, lo:hi]
+        in_splits = [max(0, min(hi, (j + 1) * shard) - max(lo, j * shard)) for j in range(world)]
+        mlo, mhi = max(lo, rank * shard), min(hi, (rank + 1) * shard)
+        m = max(0, mhi - mlo)
+        tmp = torch.empty(world * m, dtype=self.comm_dtype, device=payload.device)
+        w = self.comm.all_to_all_bucket(src, tmp, in_splits, m)
+        if w is not None:
+            self._works.append(w)
+        if m > 0:
+            self._pending_copies.append((tmp.view(world, m), row, mlo - rank * shard, mhi - rank * shard))
+
+    def mark_row_started(self, row: int) -> None:
         self._started.add(row)
 
     def exchanged(self, payload: torch.Tensor) -> torch.Tensor:
@@ -81,6 +116,9 @@ class Aggregator:
             w.wait()
         self._works = []
         self._started = set()
+        for tmp2d, row, slo, shi in self._pending_copies:
+            self._recv[row, :, slo:shi].copy_(tmp2d)
+        self._pending_copies = []
         if self.comm_dtype != torch.float32:
             self._recv32.copy_(self._recv)  # upcast once; decode kernels stay fp32
             return self._recv32.view(rows * self.comm.world, self.space.shard)
@@ -105,6 +143,7 @@ class MeanAggregator(Aggregator):
         super().__init__(comm, space)
         self.num_workers = num_workers
         self._rs_shards = None  # (rows, shard) per-row reduce_scatter outputs
+        self._bucketed = False
 
     def start_row(self, payload: torch.Tensor, row: int) -> None:
         if not self.comm.distributed:
@@ -117,9 +156,29 @@ class MeanAggregator(Aggregator):
             self._works.append(w)
         self._started.add(row)
 
+    def start_bucket(self, payload: torch.Tensor, row: int, lo: int, hi: int) -> None:
+        """Mean commutes with summing in-flight: the bucket is an in-place async
+        all_reduce of the payload slice (the hook guarantees autograd is done with
+        the range), so aggregate() just rescales the row."""
+        if not self.comm.distributed:
+            return
+        w = self.comm.all_reduce(payload[row, lo:hi], async_op=True)
+        if w is not None:
+            self._works.append(w)
+        self._bucketed = True
+
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
         if not self.comm.distributed:
             ops.sum_rows(payload, self._out)
+            self._out /= float(self.num_workers)
+            return self._out
+        if self._bucketed:
+            for w in self._works:
+                w.wait()
+            self._works = []
+            self._started = set()
+            self._bucketed = False
+            ops.sum_rows(payload, self._out)  # rows now hold cross-rank sums
             self._out /= float(self.num_workers)
             return self._out
         for r in range(payload.shape[0]):

@@ -112,12 +112,25 @@ class Communicator:
             return
         dist.all_gather_into_tensor(out, shard.contiguous())
 
-    def all_reduce(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
+    def all_reduce(self, t: torch.Tensor, op: str = "sum", async_op: bool = False):
         if not self.distributed:
-            return t
+            return t if not async_op else None
         red = {"sum": dist.ReduceOp.SUM, "min": dist.ReduceOp.MIN, "max": dist.ReduceOp.MAX}[op]
-        dist.all_reduce(t, op=red)
-        return t
+        work = dist.all_reduce(t, op=red, async_op=async_op)
+        return work if async_op else t
+
+    def all_to_all_bucket(self, send: torch.Tensor, recv_flat: torch.Tensor,
+                          in_splits: list, out_split: int):
+        """Unequal-split all_to_all for a bucket [lo, hi) of a payload row: rank j
+        receives every rank's intersection of the bucket with j's shard chunk.
+        send: the (hi-lo,) slice; recv_flat: (world*out_split,).  Always async (the
+        per-layer overlap path posts these from backward hooks)."""
+        return dist.all_to_all_single(
+            recv_flat, send,
+            output_split_sizes=[out_split] * self.world,
+            input_split_sizes=in_splits,
+            async_op=True,
+        )
 
     def broadcast(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
         if self.distributed:

@@ -90,6 +90,32 @@ class FlatSpace:
         for p in self.params:
             p.grad = None
 
+    # ------------------------------------------------------------------ buckets
+    def build_buckets(self, bucket_mb: float):
+        """Partition the parameter list into contiguous flat ranges of ~bucket_mb,
+        grouped in REVERSE parameter order (the approximate autograd fire order),
+        for the per-layer comm/compute overlap (post-accumulate-grad hooks launch
+        each bucket's exchange while the rest of backward computes).
+
+        Returns [(lo, hi, (param_indices...)), ...] in expected fire order.
+        """
+        limit = max(int(bucket_mb * 1024 * 1024 / 4), 1)
+        groups, cur, cur_elems = [], [], 0
+        for idx in reversed(range(len(self.params))):
+            cur.append(idx)
+            cur_elems += self.numels[idx]
+            if cur_elems >= limit:
+                groups.append(cur)
+                cur, cur_elems = [], 0
+        if cur:
+            groups.append(cur)
+        out = []
+        for group in groups:
+            lo = min(self.offsets[i] for i in group)
+            hi = max(self.offsets[i] + self.numels[i] for i in group)
+            out.append((lo, hi, tuple(group)))
+        return out
+
     # ------------------------------------------------------------------ shards
     def local_seg_bounds(self, rank: int) -> torch.Tensor:
         """Segment bounds clipped to this rank's shard, in shard-local coordinates.

@@ -164,6 +164,28 @@ class Trainer:
         self.use_graphs = bool(cfg.hip_graphs) and device.type == "cuda" and not cfg.deterministic
         self._graphs = {}
         self._streams = []
+        # ---------------- per-layer (bucketed) comm/compute overlap --------------
+        # Eager path only: python hooks do not fire inside a hipGraph replay, and
+        # the graph path already overlaps across logical workers' streams.  With
+        # L=1 per rank (baseline at any N, maj_vote/cyclic at N=P) this is the only
+        # mechanism that starts the gradient exchange BEFORE backward finishes —
+        # the reference's signature per-layer interleaving (lenet.py:114-218,
+        # resnet_split.py:431-623), rebuilt as post-accumulate-grad hooks.
+        self.use_buckets = (cfg.bucket_mb > 0 and not self.use_graphs
+                            and approach in ("baseline", "maj_vote")
+                            and self.comm.distributed
+                            and hasattr(self.agg, "start_bucket"))
+        if self.use_buckets:
+            self._buckets = self.space.build_buckets(cfg.bucket_mb)
+            self._param_bucket = {}
+            for bi, (_, _, idxs) in enumerate(self._buckets):
+                for i in idxs:
+                    self._param_bucket[i] = bi
+            self._bucket_left: list = []
+            self._hook_row = None
+            self._hook_adversary = False
+            for i, p in enumerate(self.space.params):
+                p.register_post_accumulate_grad_hook(self._make_bucket_hook(i))
         self.step_sync = True     # per-step device sync (bench may disable; driver
                                   # brackets with its own barrier+synchronize)
         self.collect_loss = True  # read losses to host each step
@@ -226,6 +248,42 @@ class Trainer:
             loss = self.criterion(logits, y)
         loss.backward()
         return loss.detach()
+
+    # ------------------------------------------------------------ bucketed overlap
+    def _make_bucket_hook(self, pidx: int):
+        bi = self._param_bucket[pidx]
+
+        def hook(_param):
+            if self._hook_row is None:  # not inside a bucketed backward (e.g. eval)
+                return
+            self._bucket_left[bi] -= 1
+            if self._bucket_left[bi] == 0:
+                self._ship_bucket(bi)
+
+        return hook
+
+    def _ship_bucket(self, bi: int) -> None:
+        lo, hi, _ = self._buckets[bi]
+        row = self._hook_row
+        if self._hook_adversary:
+            # injection at the send boundary, per chunk — same placement as the
+            # reference's per-layer err_simulation inside backward (lenet.py:129-141)
+            ops.inject_(self.payload[row, lo:hi], self.cfg.err_mode, cyclic=False)
+        self.agg.start_bucket(self.payload, row, lo, hi)
+
+    def _begin_bucketed_row(self, row: int, adversary: bool) -> None:
+        self._bucket_left = [len(b[2]) for b in self._buckets]
+        self._hook_row = row
+        self._hook_adversary = adversary
+
+    def _end_bucketed_row(self) -> None:
+        # flush buckets whose params never fired (unused-in-forward params keep
+        # zero grads; every rank has the identical model so flush order matches)
+        for bi, left in enumerate(self._bucket_left):
+            if left > 0:
+                self._ship_bucket(bi)
+        self.agg.mark_row_started(self._hook_row)
+        self._hook_row = None
 
     # ------------------------------------------------------------ hipGraph capture
     # ResNet-18/CIFAR is launch-bound on MI355X (~1.4k kernel dispatches per bench
@@ -337,6 +395,13 @@ class Trainer:
                     group = (self.rank - l) % self.world
                     worker_id = l * self.world + self.rank  # l-major global worker id
                 x, y = self.data.batch_for(group, step)
+                if self.use_buckets:
+                    # per-layer overlap: hooks ship ~bucket_mb chunks of this row
+                    # (inject-at-send included) while backward still computes
+                    self._begin_bucketed_row(l, worker_id in adversaries)
+                    losses.append(self._forward_backward(x, y, self.payload[l]))
+                    self._end_bucketed_row()
+                    continue
                 st = self._worker_stream(l)
                 losses.append(self._run_fwd_bwd(("slot", l), self.payload[l], x, y, stream=st))
                 pending.append((l, worker_id, st))
