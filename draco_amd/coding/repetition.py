@@ -42,6 +42,30 @@ def majority_vote_index(eq: np.ndarray) -> int:
     return maj
 
 
+def colocated_member_rows(n_groups: int, group_size: int, world0: int, alive):
+    """Recv-row coordinates of every group member over a (possibly reduced) rank set.
+
+    Member i of group g is hosted at ORIGINAL rank (g + i) % world0 (the colocated
+    layout, which never changes); after the all_to_all over the survivor group its
+    recv row is i * W' + pos(host) with W' = len(alive).  Members hosted on dead
+    ranks are forfeited (mask False) — the erasure form of the repetition decode.
+
+    Returns (member_rows (G, r) int64, member_mask (G, r) bool).
+    """
+    pos = {rk: i for i, rk in enumerate(alive)}
+    Wp = max(len(alive), 1)
+    G, r = n_groups, group_size
+    rows = np.zeros((G, r), dtype=np.int64)
+    mask = np.zeros((G, r), dtype=bool)
+    for g in range(G):
+        for i in range(r):
+            h = (g + i) % world0
+            if h in pos:
+                rows[g, i] = i * Wp + pos[h]
+                mask[g, i] = True
+    return rows, mask
+
+
 def group_membership(n_groups: int, group_size: int, world: int):
     """Member i of group g is hosted as local worker i of rank (g + i) % world.
 

@@ -52,6 +52,9 @@ class Config:
     bucket_mb: float = 25.0       # per-layer overlap: backward hooks ship the payload
                                   # row in ~bucket_mb chunks while backward continues
                                   # (eager path, baseline/maj_vote; 0 = whole-row)
+    health_timeout: float = 0.0   # colocated topology: heartbeat staleness (seconds)
+                                  # after which a rank is declared dead and its
+                                  # logical workers become erasures (0 = disabled)
     nan_guard: bool = True        # failure detection: skip updates on non-finite decode
     gpu_timing: bool = False      # device-accurate phase spans via HIP events (metrics)
     compile: bool = False         # torch.compile the model before hipGraph capture

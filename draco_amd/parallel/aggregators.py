@@ -217,7 +217,8 @@ class VoteAggregator(Aggregator):
     name = "maj_vote"
 
     def __init__(self, comm, space, group_size: int, atol: float = 0.0,
-                 rtol: float = 0.0, member_rows=None, comm_dtype=torch.float32):
+                 rtol: float = 0.0, member_rows=None, comm_dtype=torch.float32,
+                 member_mask=None):
         super().__init__(comm, space, comm_dtype)
         self.atol = atol
         self.rtol = rtol
@@ -231,25 +232,37 @@ class VoteAggregator(Aggregator):
             )
         self.member_rows = np.asarray(member_rows)  # (G, r)
         self.G, self.r = self.member_rows.shape
-        pairs_a, pairs_b = [], []
-        for rows in self.member_rows:
+        # member_mask[g, i] False = forfeited member (its host rank died; the vote
+        # proceeds over the remaining members — the repetition code's erasure form)
+        if member_mask is None:
+            member_mask = np.ones((self.G, self.r), dtype=bool)
+        self.member_mask = np.asarray(member_mask, dtype=bool)
+        pairs_a, pairs_b, pg, pi, pj = [], [], [], [], []
+        for g, rows in enumerate(self.member_rows):
             for i in range(self.r):
                 for j in range(i + 1, self.r):
-                    pairs_a.append(rows[i])
-                    pairs_b.append(rows[j])
+                    if self.member_mask[g, i] and self.member_mask[g, j]:
+                        pairs_a.append(rows[i])
+                        pairs_b.append(rows[j])
+                        pg.append(g)
+                        pi.append(i)
+                        pj.append(j)
         self.pairs_a = torch.tensor(pairs_a, dtype=torch.int64, device=space.device)
         self.pairs_b = torch.tensor(pairs_b, dtype=torch.int64, device=space.device)
-        self.n_pairs_per_group = self.r * (self.r - 1) // 2
         # device-side winner selection state (no host round-trip in aggregate():
         # at N=8 a per-step .to("cpu") serialises all ranks on one readback)
         dev = space.device
-        npg = self.n_pairs_per_group
-        self._pg = torch.arange(self.G, device=dev).repeat_interleave(npg)
-        ij = [(i, j) for i in range(self.r) for j in range(i + 1, self.r)]
-        self._pi = torch.tensor([i for i, _ in ij] * self.G, dtype=torch.int64, device=dev)
-        self._pj = torch.tensor([j for _, j in ij] * self.G, dtype=torch.int64, device=dev)
-        self._eye = torch.eye(self.r, device=dev).expand(self.G, self.r, self.r).contiguous()
+        self._pg = torch.tensor(pg, dtype=torch.int64, device=dev)
+        self._pi = torch.tensor(pi, dtype=torch.int64, device=dev)
+        self._pj = torch.tensor(pj, dtype=torch.int64, device=dev)
+        eye = torch.eye(self.r, device=dev).expand(self.G, self.r, self.r).clone()
+        maskt = torch.tensor(self.member_mask, device=dev)
+        eye *= maskt[:, :, None].float()  # forfeited members have empty classes
+        self._eye = eye
         self._eqm = torch.empty_like(self._eye)
+        # forfeited members can never win the argmax
+        self._class_bias = torch.where(maskt, 0.0, -1e9)
+        self._alive_count = maskt.sum(dim=1).float()  # (G,)
         self._member_rows_t = torch.tensor(self.member_rows, dtype=torch.int64, device=dev)
         # telemetry: steps where some group saw NO equal pair (tolerance too tight /
         # replicas diverged -> the vote degenerates to an arbitrary pick, which an
@@ -290,11 +303,13 @@ class VoteAggregator(Aggregator):
         eqm.copy_(self._eye)
         eqm[self._pg, self._pi, self._pj] = eqf
         eqm[self._pg, self._pj, self._pi] = eqf
-        class_size = eqm.sum(dim=2)  # (G, r)
-        winner_member = class_size.argmax(dim=1)  # first-max tie-break: deterministic
+        class_size = eqm.sum(dim=2)  # (G, r); forfeited members contribute/score 0
+        winner_member = (class_size + self._class_bias).argmax(dim=1)  # deterministic tie-break
         winners = self._member_rows_t.gather(1, winner_member.unsqueeze(1)).squeeze(1)
         if self.r > 1:
-            self._deg_counter += (class_size.max(dim=1).values <= 1.5).any().to(torch.int64)
+            # degenerate: a group with >= 2 alive members and NO equal pair
+            deg = (self._alive_count >= 2) & (class_size.max(dim=1).values <= 1.5)
+            self._deg_counter += deg.any().to(torch.int64)
         ops.mean_rows(recv, winners, self._shard_out)
         self.comm.all_gather_shard(self._shard_out, self._out)
         return self._out
@@ -414,25 +429,39 @@ class CyclicAggregator(Aggregator):
     name = "cyclic"
 
     def __init__(self, comm, space, code: CyclicCode, workers_per_rank: int,
-                 comm_dtype=torch.float32):
+                 comm_dtype=torch.float32, world0: int | None = None, alive=None):
         super().__init__(comm, space, comm_dtype)
         self.code = code
         self.L = workers_per_rank
         self.n = code.n
-        world = max(comm.world, 1)
-        # worker w = l*world + src; its (re, im) payload rows at the source are
-        # (2l, 2l+1), so after the l-major all_to_all its recv rows are
-        # (2l+p)*world + src
+        # Logical worker w = l*world0 + src0 where world0 is the ORIGINAL world size
+        # (the code is built for n = L*world0 and never changes).  `alive` lists the
+        # surviving original ranks; workers hosted on dead ranks are permanent
+        # ERASURES — known-bad rows the decode removes (<= s of them total with the
+        # step's adversaries).  Recv rows are in survivor-world coordinates:
+        # worker w's (re, im) payload rows (2l, 2l+1) land at (2l+p)*W' + pos(src0).
+        world0 = world0 if world0 is not None else max(comm.world, 1)
+        alive = list(alive) if alive is not None else list(range(world0))
+        pos = {rk: i for i, rk in enumerate(alive)}
+        Wp = max(len(alive), 1)
         w_ids = np.arange(self.n)
-        l, src = w_ids // world, w_ids % world
-        self.rows_re = (2 * l) * world + src
-        self.rows_im = (2 * l + 1) * world + src
+        l, src = w_ids // world0, w_ids % world0
+        self.alive_w = np.array([int(s) in pos for s in src])
+        posv = np.array([pos.get(int(s), 0) for s in src])
+        self.rows_re = (2 * l) * Wp + posv
+        self.rows_im = (2 * l + 1) * Wp + posv
+        self.erased = frozenset(int(w) for w in w_ids[~self.alive_w])
+        aw = w_ids[self.alive_w]
+        self._alive_rows = torch.tensor(
+            np.concatenate([self.rows_re[aw], self.rows_im[aw]]),
+            dtype=torch.int64, device=space.device)
+        self._alive_idx = aw
         self._z = None
         self._zgen = None
 
     def aggregate(self, payload_planes: torch.Tensor, step: int,
                   erasures: frozenset = frozenset()) -> torch.Tensor:
-        recv = self.exchanged(payload_planes)  # (2L*world, shard)
+        recv = self.exchanged(payload_planes)  # (2L*world', shard)
         # random projection generated ON DEVICE (a shard-sized CPU randn + H2D copy
         # costs ~30 ms at d=11M — it dominated the whole decode); per-rank streams
         # may legitimately use different z (partial projections are summed)
@@ -441,22 +470,23 @@ class CyclicAggregator(Aggregator):
             self._zgen = torch.Generator(device=self.space.device)
         self._zgen.manual_seed(0x5EED ^ (step * 1000003) ^ self.comm.rank)
         z = self._z.normal_(mean=1.0, std=1.0, generator=self._zgen)
-        proj = ops.cyclic_project(recv, z)  # (2L*world,) per-row partial dots
+        proj = ops.cyclic_project(recv, z)  # (2L*world',) per-row partial dots
         self.comm.all_reduce(proj)
         pa = proj.to("cpu").numpy().astype(np.float64)
-        proj_complex = pa[self.rows_re] + 1j * pa[self.rows_im]
+        proj_complex = np.where(self.alive_w, pa[self.rows_re] + 1j * pa[self.rows_im], 0.0)
+        known_bad = frozenset(erasures) | self.erased
         syndrome = self.code.W_perp @ proj_complex
         scale = float(np.abs(proj_complex).max())
-        if not erasures and float(np.abs(syndrome).max()) <= 1e-7 * max(scale, 1e-30):
+        if not known_bad and float(np.abs(syndrome).max()) <= 1e-7 * max(scale, 1e-30):
             healthy = np.arange(self.n)
         else:
-            healthy = self.code.locate_errors(syndrome, known_bad=erasures)
+            healthy = self.code.locate_errors(syndrome, known_bad=known_bad)
         v = self.code.recombination_vector(healthy)
         # Re(v @ R) = sum_w vre[w]*re_row[w] - vim[w]*im_row[w]: one combine kernel
-        rows = torch.tensor(np.concatenate([self.rows_re, self.rows_im]),
-                            dtype=torch.int64, device=recv.device)
-        w = torch.tensor(np.concatenate([np.real(v), -np.imag(v)]) / self.n,
+        # over the ALIVE workers' rows (v is zero on erased/adversarial rows anyway)
+        va = v[self._alive_idx]
+        w = torch.tensor(np.concatenate([np.real(va), -np.imag(va)]) / self.n,
                          dtype=torch.float32, device=recv.device)
-        ops.combine_rows(recv, rows, w, self._shard_out)
+        ops.combine_rows(recv, self._alive_rows, w, self._shard_out)
         self.comm.all_gather_shard(self._shard_out, self._out)
         return self._out

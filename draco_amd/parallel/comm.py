@@ -21,11 +21,13 @@ import torch.distributed as dist
 
 
 class Communicator:
-    def __init__(self, rank: int, world: int, device: torch.device, backend: str | None = None):
+    def __init__(self, rank: int, world: int, device: torch.device, backend: str | None = None,
+                 group=None):
         self.rank = rank
         self.world = world
         self.device = device
         self.distributed = world > 1
+        self.group = group  # None = default process group; else a dist subgroup
         if self.distributed and not dist.is_initialized():
             if backend is None:
                 backend = "nccl" if device.type == "cuda" else "gloo"
@@ -41,6 +43,19 @@ class Communicator:
             )
         self.backend = dist.get_backend() if self.distributed else "local"
 
+    def subgroup_communicator(self, group, ranks: list) -> "Communicator":
+        """A Communicator bound to a pre-created dist subgroup (survivor set after a
+        rank failure — parallel/health.py).  rank/world are POSITIONAL within the
+        subgroup; collectives address only the surviving ranks."""
+        c = Communicator.__new__(Communicator)
+        c.device = self.device
+        c.group = group
+        c.world = len(ranks)
+        c.rank = ranks.index(dist.get_rank()) if c.world > 1 else 0
+        c.distributed = c.world > 1
+        c.backend = self.backend
+        return c
+
     @classmethod
     def from_env(cls, device: torch.device, backend: str | None = None) -> "Communicator":
         rank = int(os.environ.get("RANK", 0))
@@ -51,9 +66,9 @@ class Communicator:
     def barrier(self):
         if self.distributed:
             if self.backend == "nccl":
-                dist.barrier(device_ids=[self.device.index])
+                dist.barrier(device_ids=[self.device.index], group=self.group)
             else:
-                dist.barrier()
+                dist.barrier(group=self.group)
 
     def all_to_all_row(self, payload_row: torch.Tensor, recv_row: torch.Tensor,
                        async_op: bool = False):
@@ -65,7 +80,7 @@ class Communicator:
         if not self.distributed:
             recv_row.view(-1).copy_(payload_row)
             return None
-        work = dist.all_to_all_single(recv_row.view(-1), payload_row, async_op=async_op)
+        work = dist.all_to_all_single(recv_row.view(-1), payload_row, group=self.group, async_op=async_op)
         return work if async_op else None
 
     def all_to_all_rows(self, payload: torch.Tensor) -> torch.Tensor:
@@ -93,7 +108,7 @@ class Communicator:
         if not self.distributed:
             out_shard.copy_(payload_row)
             return None
-        work = dist.reduce_scatter_tensor(out_shard, payload_row, async_op=async_op)
+        work = dist.reduce_scatter_tensor(out_shard, payload_row, group=self.group, async_op=async_op)
         return work if async_op else None
 
     def reduce_scatter_sum(self, payload_sum: torch.Tensor) -> torch.Tensor:
@@ -102,7 +117,7 @@ class Communicator:
         if not self.distributed:
             return payload_sum
         out = torch.empty(shard, dtype=payload_sum.dtype, device=payload_sum.device)
-        dist.reduce_scatter_tensor(out, payload_sum)
+        dist.reduce_scatter_tensor(out, payload_sum, group=self.group)
         return out
 
     def all_gather_shard(self, shard: torch.Tensor, out: torch.Tensor) -> None:
@@ -110,13 +125,13 @@ class Communicator:
         if not self.distributed:
             out.copy_(shard)
             return
-        dist.all_gather_into_tensor(out, shard.contiguous())
+        dist.all_gather_into_tensor(out, shard.contiguous(), group=self.group)
 
     def all_reduce(self, t: torch.Tensor, op: str = "sum", async_op: bool = False):
         if not self.distributed:
             return t if not async_op else None
         red = {"sum": dist.ReduceOp.SUM, "min": dist.ReduceOp.MIN, "max": dist.ReduceOp.MAX}[op]
-        work = dist.all_reduce(t, op=red, async_op=async_op)
+        work = dist.all_reduce(t, op=red, group=self.group, async_op=async_op)
         return work if async_op else t
 
     def all_to_all_bucket(self, send: torch.Tensor, recv_flat: torch.Tensor,
@@ -129,12 +144,13 @@ class Communicator:
             recv_flat, send,
             output_split_sizes=[out_split] * self.world,
             input_split_sizes=in_splits,
+            group=self.group,
             async_op=True,
         )
 
     def broadcast(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
         if self.distributed:
-            dist.broadcast(t, src=src)
+            dist.broadcast(t, src=src, group=self.group)
         return t
 
     # ------------------------------------------------------------------ p2p (PS mode)

@@ -89,15 +89,25 @@ class Trainer:
 
         self.autocast_dtype = torch.bfloat16 if (cfg.dtype == "bf16" and device.type == "cuda") else None
 
+        # ------------------ failure detection (colocated topology) ---------------
+        # Heartbeats + pre-created survivor subgroups: a dead rank's logical workers
+        # become erasures and training continues (parallel/health.py; the reference
+        # hangs forever, baseline_master.py:112-116).  Worker identity below is
+        # defined by the ORIGINAL (rank, world) and never changes after a failure;
+        # only the comm-layer layout (self.comm.rank/world) is rebuilt.
+        self.health = None
+        if cfg.health_timeout > 0 and self.comm.distributed:
+            from .health import HealthMonitor
+
+            self.health = HealthMonitor(self.rank, self.world, cfg.health_timeout)
+
         # ---------------- logical workers & aggregator ----------------
         approach = cfg.approach
         self.approach = approach
         if approach == "baseline":
             self.L = 1
             self.P = self.world
-            self.agg = self._baseline_aggregator(cfg)
             self.data = GroupBatchSource(self._dataset(), cfg.batch_size, n_groups=self.P)
-            self.payload = self.space.alloc_payload(self.L)
         elif approach == "maj_vote":
             self.r = cfg.group_size
             self.L = self.r
@@ -121,11 +131,7 @@ class Trainer:
                 else:
                     rtol = 1e-4  # fp32 reorder noise
             self.vote_rtol = rtol
-            self.agg = VoteAggregator(self.comm, self.space, group_size=self.r,
-                                      atol=cfg.vote_atol, rtol=rtol,
-                                      comm_dtype=self._comm_dtype())
             self.data = GroupBatchSource(self._dataset(), cfg.batch_size, n_groups=self.G)
-            self.payload = self.space.alloc_payload(self.L)
         elif approach == "cyclic":
             wpr = cfg.workers_per_rank or max(1, math.ceil((2 * cfg.worker_fail + 2) / self.world))
             self.L = wpr
@@ -133,10 +139,7 @@ class Trainer:
             self.P = self.n
             self.code = build_cyclic_code(self.n, cfg.worker_fail)
             self.s_hat = self.code.s_hat
-            self.agg = CyclicAggregator(self.comm, self.space, self.code, self.L,
-                                        comm_dtype=self._comm_dtype())
             self.data = GlobalBatchSource(self._dataset(), cfg.batch_size, n_workers=self.n)
-            self.payload = self.space.alloc_payload(self.L * 2)
             W = self.code.W
             sup = self.code.support
             # Sub-batch dedup within a rank: logical workers hosted on the SAME rank
@@ -150,7 +153,6 @@ class Trainer:
             needed = sorted({int(j) for w in local_w for j in sup[w]})
             self._local_subs = needed
             row_of = {j: i for i, j in enumerate(needed)}
-            self.scratch = self.space.alloc_payload(len(needed))
             self._w_re, self._w_im, self._enc_rows = [], [], []
             for l, w_global in enumerate(local_w):
                 coeff = W[w_global, sup[w_global]]
@@ -160,6 +162,7 @@ class Trainer:
                                                    dtype=torch.int64, device=device))
         else:
             raise ValueError(f"unknown approach {approach!r}")
+        self._setup_decode()
 
         self.use_graphs = bool(cfg.hip_graphs) and device.type == "cuda" and not cfg.deterministic
         self._graphs = {}
@@ -225,14 +228,92 @@ class Trainer:
         # space (the 2**40 offset in evaluate(); splitmix64 seeding cannot collide)
         return SyntheticClassification(self.cfg.dataset, self.device, seed=1234)
 
-    def _baseline_aggregator(self, cfg: Config):
+    def _baseline_aggregator(self, cfg: Config, num_workers: int):
         if cfg.mode == "normal":
-            return MeanAggregator(self.comm, self.space, num_workers=self.world)
+            return MeanAggregator(self.comm, self.space, num_workers=num_workers)
         if cfg.mode == "geometric_median":
-            return GeoMedianAggregator(self.comm, self.space, num_workers=self.world)
+            return GeoMedianAggregator(self.comm, self.space, num_workers=num_workers)
         if cfg.mode == "krum":
-            return KrumAggregator(self.comm, self.space, num_workers=self.world, s=cfg.worker_fail)
+            return KrumAggregator(self.comm, self.space, num_workers=num_workers, s=cfg.worker_fail)
         raise ValueError(f"baseline approach supports modes normal/geometric_median/krum, got {cfg.mode!r}")
+
+    def _setup_decode(self) -> None:
+        """Build the aggregator + comm-layout buffers for the CURRENT communicator
+        (called at init, and again with the survivor communicator after a rank
+        failure — worker identity keys off the ORIGINAL self.rank/self.world)."""
+        cfg = self.cfg
+        alive = self.health.alive if self.health is not None else list(range(self.world))
+        if self.approach == "baseline":
+            self.agg = self._baseline_aggregator(cfg, num_workers=len(alive))
+            self.payload = self.space.alloc_payload(self.L)
+        elif self.approach == "maj_vote":
+            from ..coding import colocated_member_rows
+
+            rows, mask = colocated_member_rows(self.G, self.r, self.world, alive)
+            self.agg = VoteAggregator(self.comm, self.space, group_size=self.r,
+                                      atol=cfg.vote_atol, rtol=self.vote_rtol,
+                                      member_rows=rows, member_mask=mask,
+                                      comm_dtype=self._comm_dtype())
+            self.payload = self.space.alloc_payload(self.L)
+        else:  # cyclic
+            self.agg = CyclicAggregator(self.comm, self.space, self.code, self.L,
+                                        comm_dtype=self._comm_dtype(),
+                                        world0=self.world, alive=alive)
+            self.payload = self.space.alloc_payload(self.L * 2)
+            self.scratch = self.space.alloc_payload(len(self._local_subs))
+
+    def _handle_failure(self, dead: list) -> None:
+        """Commit a rank failure: switch to the survivor subgroup, re-shard the flat
+        space, rebuild the decode with the dead rank's workers as erasures, and
+        adopt a canonical survivor's (params, optimizer, step) — survivors may have
+        stopped on different sides of the failed step's update, so state is
+        broadcast from the lowest surviving rank rather than assumed equal."""
+        from ..utils.checkpoint import _flat_to_params, _params_to_flat
+
+        group, alive = self.health.declare_dead(dead)
+        self.logger.log({"step": self.step_num, "event": "rank_failure",
+                         "dead": [int(r) for r in dead], "alive": alive})
+        old_space = self.space
+        stash = {}
+        for name in ("buf", "exp_avg", "exp_avg_sq", "max_exp_avg_sq"):
+            t = getattr(self.opt, name, None)
+            if isinstance(t, torch.Tensor):
+                stash[name] = _flat_to_params(old_space, t)
+        deg = getattr(self.agg, "_deg_counter", None)
+        self.comm = self.comm.subgroup_communicator(group, alive)
+        self.space = FlatSpace(self.model, self.comm.world, self.device,
+                               channels_last=self.use_cl)
+        self.opt.param = self.space.flat_param
+        for name, tensors in stash.items():
+            new = torch.zeros_like(self.space.flat_param)
+            _params_to_flat(self.space, tensors, new)
+            setattr(self.opt, name, new)
+        self._graphs = {}  # captured graphs point into the old payload/space
+        if self.use_buckets:
+            self._hook_row = None
+            self._buckets = self.space.build_buckets(self.cfg.bucket_mb)
+        self._setup_decode()
+        if deg is not None:
+            self.agg._deg_counter.copy_(deg.to(self.agg._deg_counter.device))
+        # canonical-state adoption (lowest surviving rank wins)
+        src = int(alive[0])
+        self.comm.broadcast(self.space.flat_param, src=src)
+        for name in stash:
+            self.comm.broadcast(getattr(self.opt, name), src=src)
+        bufs = [b for b in self.model.buffers() if b.dtype.is_floating_point]
+        if bufs:
+            flatb = torch.cat([b.reshape(-1).float() for b in bufs])
+            self.comm.broadcast(flatb, src=src)
+            off = 0
+            with torch.no_grad():
+                for b in bufs:
+                    n = b.numel()
+                    b.copy_(flatb[off : off + n].view(b.shape).to(b.dtype))
+                    off += n
+        meta = torch.tensor([self.step_num], dtype=torch.int64,
+                            device=self.device if self.comm.backend == "nccl" else "cpu")
+        self.comm.broadcast(meta, src=src)
+        self.step_num = int(meta[0])
 
     def _forward_backward(self, x, y, grad_row: torch.Tensor):
         if self.use_cl and x.dim() == 4:
@@ -378,6 +459,27 @@ class Trainer:
 
     # ------------------------------------------------------------------ one step
     def train_step(self) -> dict:
+        if self.health is None:
+            return self._train_step_inner()
+        self.health.beat(self.step_num)
+        dead = self.health.check()
+        if dead:
+            self._handle_failure(dead)
+        try:
+            return self._train_step_inner()
+        except RuntimeError as err:
+            # a peer dying mid-collective surfaces as a RuntimeError on the op;
+            # confirm via heartbeats, commit the failure, re-run the step (the
+            # canonical-state adoption in _handle_failure makes the retry exact
+            # whichever side of the update each survivor stopped on)
+            try:
+                dead = self.health.wait_for_dead()
+            except RuntimeError:
+                raise err  # not a failure we recognise — propagate the original
+            self._handle_failure(dead)
+            return self._train_step_inner()
+
+    def _train_step_inner(self) -> dict:
         cfg = self.cfg
         step = self.step_num
         t0 = time.perf_counter()
@@ -538,7 +640,7 @@ class Trainer:
             return
         flat = torch.cat([b.reshape(-1).float() for b in bufs])
         self.comm.all_reduce(flat)
-        flat /= self.world
+        flat /= self.comm.world  # survivors only, after a failure
         off = 0
         with torch.no_grad():
             for b in bufs:
@@ -556,4 +658,6 @@ class Trainer:
         self.step_num = step
 
     def close(self):
+        if self.health is not None:
+            self.health.close()
         self.logger.close()

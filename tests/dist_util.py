@@ -28,9 +28,10 @@ def _entry(fn, rank, world, port, q, args, kwargs):
         q.put((rank, "err", traceback.format_exc()))
 
 
-def run_dist(fn, world: int, *args, timeout: float = 180.0, **kwargs):
+def run_dist(fn, world: int, *args, timeout: float = 180.0, expect_missing=(), **kwargs):
     """Run fn(rank, world, *args) in `world` processes over gloo; returns
-    {rank: result}.  Raises on any child failure."""
+    {rank: result}.  Raises on any child failure.  Ranks in `expect_missing` are
+    allowed to die without reporting (failure-injection tests)."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     port = _free_port()
@@ -42,7 +43,7 @@ def run_dist(fn, world: int, *args, timeout: float = 180.0, **kwargs):
         p.start()
     results = {}
     errs = []
-    for _ in range(world):
+    for _ in range(world - len(set(expect_missing))):
         rank, status, payload = q.get(timeout=timeout)
         if status == "ok":
             results[rank] = payload

@@ -1,0 +1,73 @@
+"""Rank-failure recovery in the colocated topology (gloo, CPU).
+
+A killed rank's logical workers become erasures: the vote drops the forfeited
+members, the cyclic decode removes the known-bad rows, and training continues on
+the survivor subgroup with bit-identical params across survivors.  (The reference
+hangs forever on a dead worker — baseline_master.py:112-116.)
+
+Capability note: erasures spend the same redundancy budget as adversaries
+(repetition: forfeits+adversaries < r/2 per group; cyclic: erased+adversarial
+rows <= s), so these tests run with err_mode="none" — one dead rank IS the fault.
+Rank 0 hosts the rendezvous TCPStore, so rank 0 death is unrecoverable by design;
+tests kill a non-zero rank.
+"""
+import os
+import time
+
+import pytest
+import torch
+
+from tests.dist_util import run_dist
+
+pytestmark = pytest.mark.timeout(300)
+
+
+def _failure_worker(rank, world, approach, kw, die_rank, die_step):
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                 approach=approach, err_mode="none", max_steps=100, eval_freq=0,
+                 log_dir="", train_dir="/tmp/draco_fail", health_timeout=1.5, **kw)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    losses = []
+    for i in range(12):
+        if rank == die_rank and i == die_step:
+            # abrupt death mid-run: report to the harness, then hard-exit
+            time.sleep(0.2)
+            os._exit(0)
+        losses.append(t.train_step()["loss"])
+    h = float(t.space.flat_param.double().sum())
+    alive = list(t.health.alive)
+    t.close()
+    return (losses, h, alive)
+
+
+def _run_failure(world, approach, kw, die_rank=None):
+    die_rank = world - 1 if die_rank is None else die_rank
+    res = run_dist(_failure_worker, world, approach, kw, die_rank, 3,
+                   timeout=240.0, expect_missing={die_rank})
+    survivors = [r for r in range(world) if r != die_rank]
+    hashes = [res[r][1] for r in survivors]
+    assert all(h == hashes[0] for h in hashes), "survivor params diverged"
+    for r in survivors:
+        assert res[r][2] == survivors, f"rank {r} alive set wrong: {res[r][2]}"
+        losses = res[r][0]
+        assert len(losses) == 12
+        assert losses[-1] < losses[0], "training did not keep converging after the failure"
+    return res
+
+
+def test_vote_survives_rank_death():
+    _run_failure(3, "maj_vote", dict(mode="maj_vote", group_size=3, worker_fail=1))
+
+
+def test_cyclic_survives_rank_death():
+    # world=3, L=2 -> n=6 workers, s=2: killing one rank erases its 2 workers,
+    # within the decode's known-bad budget (<= s)
+    _run_failure(3, "cyclic", dict(mode="cyclic", worker_fail=2, workers_per_rank=2))
+
+
+def test_baseline_mean_survives_rank_death():
+    _run_failure(3, "baseline", dict(mode="normal", worker_fail=0))
