@@ -74,6 +74,23 @@ class _PSBase:
         self.P = self.world - 1
         if self.P < 1:
             raise RuntimeError("ps topology needs world_size >= 2")
+        if self.comm.backend == "nccl" and os.environ.get("DRACO_PS_NCCL_OK") != "1":
+            # refuse loudly rather than fail subtly: the PS lane's straggler
+            # timeout relies on host-blocking p2p waits (gloo semantics) and the
+            # mode has not been validated on RCCL hardware.  The colocated
+            # topology (trainer.py) is the GPU path.  Set DRACO_PS_NCCL_OK=1 to
+            # run it anyway (plain broadcast+send/recv are RCCL-supported).
+            raise RuntimeError(
+                "topology=ps has not been validated on the nccl/RCCL backend; "
+                "use topology=colocated on GPUs or set DRACO_PS_NCCL_OK=1")
+        # backend-independent out-of-band signalling (mid-step preemption/abort):
+        # tagged p2p is NOT honored by NCCL (tags are ignored -> mis-matched recvs,
+        # not errors), so the abort channel lives in the rendezvous TCPStore
+        import torch.distributed as dist
+
+        self._store = dist.distributed_c10d._get_default_store()
+        if self.rank == 0:
+            self._store.set("ps_abort", "0")
 
         torch.manual_seed(cfg.seed)
         self.model = build_model(cfg.network, cfg.dataset).to(device)
@@ -141,13 +158,15 @@ class Master(_PSBase):
     def request_abort(self):
         """Preempt workers MID-STEP (the reference's tag-77 kill channel,
         lenet.py:237-240 / resnet_split.py:636-640 — which had no master-side sender;
-        this is it).  Workers abandon remaining redundant sub-batches, send what they
-        have, and exit at the next step boundary."""
+        this is it).  Delivered through the TCPStore, not tagged p2p: NCCL silently
+        ignores tags (a mis-matched recv, not an exception, would be the failure
+        mode) and a pending p2p recv wedges gloo teardown.  Workers poll the flag
+        between redundant sub-batches, ship what they have, and exit at the next
+        step boundary."""
         if self._abort:
             return
         self._abort = True
-        self._abort_reqs = [self.comm.isend(torch.zeros(1), dst=w + 1, tag=77)
-                            for w in range(self.P)]
+        self._store.set("ps_abort", "1")
 
     def run(self, max_steps: int | None = None):
         cfg = self.cfg
@@ -192,12 +211,8 @@ class Master(_PSBase):
         ctrl[0] = self.step_num
         ctrl[1] = 1
         self.comm.broadcast(ctrl, src=0)
-        # complete every worker's armed tag-77 preemption irecv — a pending p2p recv
-        # makes the gloo context abort at process teardown
         if not self._abort:
-            self.request_abort()
-        for r in getattr(self, "_abort_reqs", []):
-            r.wait()
+            self.request_abort()  # store flag: lets worker poll threads exit
         self.logger.close()
 
     def _gather_grads(self) -> frozenset:
@@ -298,25 +313,26 @@ class Worker(_PSBase):
         self._preempt = self._arm_preempt()
 
     def _arm_preempt(self):
-        """Out-of-band preemption: tag-77 irecv from the master, observed through a
-        daemon waiter thread (gloo Work.is_completed never fires for p2p recvs
-        without wait())."""
+        """Out-of-band preemption: a daemon thread polls the master's store-side
+        abort flag (backend-independent — tagged p2p would be silently broken on
+        NCCL) and sets a local event the compute loop checks between redundant
+        sub-batches."""
         import threading
 
         ev = threading.Event()
-        buf = torch.zeros(1)
-        try:
-            req = self.comm.irecv(buf, src=0, tag=77)
-        except Exception:
-            return ev  # backend without tagged p2p: preemption disabled
+        store = self._store
 
-        def waiter():
-            try:
-                req.wait()
-            finally:
-                ev.set()
+        def poller():
+            while not ev.is_set():
+                try:
+                    if store.get("ps_abort").decode() == "1":
+                        ev.set()
+                        return
+                except Exception:
+                    return  # store gone: process is shutting down
+                time.sleep(0.05)
 
-        th = threading.Thread(target=waiter, daemon=True)
+        th = threading.Thread(target=poller, daemon=True)
         th.start()
         self._preempt_thread = th
         return ev
