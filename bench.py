@@ -94,38 +94,61 @@ def main():
     t.logger.close()
     on_gpu = t.device.type == "cuda"
 
-    for _ in range(args.warmup):
-        t.train_step()
-    # inside the timed region: no per-step host sync, no per-step loss readback
-    # (the bracket below barriers + synchronizes; every step's work still runs fully)
-    t.step_sync = False
-    t.collect_loss = False
+    def timed_region(tr, steps, warmup):
+        """warmup untimed, then EXACTLY `steps` steps bracketed by barrier+sync on
+        both sides; returns MAX elapsed over ranks (slowest rank defines the job)."""
+        for _ in range(warmup):
+            tr.train_step()
+        # inside the timed region: no per-step host sync, no per-step loss readback
+        # (the bracket barriers + synchronizes; every step's work still runs fully)
+        tr.step_sync = False
+        tr.collect_loss = False
+        tr.comm.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            tr.train_step()
+        if on_gpu:
+            torch.cuda.synchronize()
+        tr.comm.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+        elapsed = time.perf_counter() - t0
+        tr.step_sync = True
+        tr.collect_loss = True
+        el = torch.tensor([elapsed], dtype=torch.float64)
+        if tr.comm.distributed:
+            el_dev = el.to(tr.device) if tr.comm.backend == "nccl" else el
+            tr.comm.all_reduce(el_dev, op="max")
+            elapsed = float(el_dev[0])
+        return elapsed
 
-    t.comm.barrier()
-    if on_gpu:
-        torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    for _ in range(args.steps):
-        t.train_step()
-    if on_gpu:
-        torch.cuda.synchronize()
-    t.comm.barrier()
-    if on_gpu:
-        torch.cuda.synchronize()
-    elapsed = time.perf_counter() - t0
-
-    # MAX over ranks (slowest rank defines the job step time)
-    el = torch.tensor([elapsed], dtype=torch.float64)
-    if t.comm.distributed:
-        el_dev = el.to(t.device) if t.comm.backend == "nccl" else el
-        t.comm.all_reduce(el_dev, op="max")
-        elapsed = float(el_dev[0])
+    elapsed = timed_region(t, args.steps, args.warmup)
 
     # post-bracket honesty telemetry: one more step with loss readback
-    t.collect_loss = True
-    t.step_sync = True
     final_loss = t.train_step()["loss"]
     ms_per_step = elapsed / args.steps * 1000.0
+
+    # ---- plain-DP reference arm: same model/batch/world, approach=baseline mode=
+    # normal, NO adversary — quantifies the coding overhead on the same box.
+    # vs_baseline = coded images/sec ÷ plain-DP images/sec (the reference repo
+    # publishes no numbers, BASELINE.md, so the plain arm is the denominator).
+    vs_baseline = None
+    baseline_ips = None
+    if args.approach != "baseline":
+        import dataclasses
+
+        bcfg = dataclasses.replace(
+            cfg, approach="baseline", mode="normal", worker_fail=0, err_mode="none",
+            train_dir="gpurun_out/bench_ckpt_base")
+        tb = Trainer(bcfg)
+        tb.logger.stdout_every = 0
+        tb.logger.close()
+        bsteps = max(min(args.steps, 10), 1)
+        belapsed = timed_region(tb, bsteps, min(args.warmup, 3))
+        baseline_ips = world * args.batch_size * bsteps / belapsed
+        tb.close()
     # DISTINCT images per step (redundant compute is the price of the code and is
     # not counted): maj_vote -> G*B (G=world groups); cyclic -> n*B global batch;
     # baseline -> world*B
@@ -141,6 +164,8 @@ def main():
         distinct_per_step = world * args.batch_size
         processed = distinct_per_step
     images_per_sec = distinct_per_step * args.steps / elapsed
+    if baseline_ips is not None:
+        vs_baseline = round(images_per_sec / baseline_ips, 4)
 
     # restore the real stdout for the contract line
     sys.stdout.flush()
@@ -159,7 +184,10 @@ def main():
             "ms_per_step": round(ms_per_step, 3),
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": None,
+            # coded throughput ÷ plain-DP (baseline normal, no adversary) on the
+            # same box — the reference publishes no absolute numbers
+            "vs_baseline": vs_baseline,
+            "baseline_images_per_sec": round(baseline_ips, 2) if baseline_ips else None,
             "final_loss": round(final_loss, 4) if final_loss == final_loss else None,
             "skipped_updates": t.skipped_updates,
             "vote_degenerate_steps": getattr(t.agg, "degenerate_steps", None),
