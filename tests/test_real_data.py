@@ -103,3 +103,40 @@ def test_trainer_uses_real_data(tmp_path):
     rec = t.train_step()
     assert np.isfinite(rec["loss"])
     t.close()
+
+
+def test_cifar_augmentation(tmp_path):
+    """RandomCrop(32, pad=4) + flip (reference util.py:29-65): deterministic per
+    start index, shape-preserving, and every output is a genuine crop/flip of the
+    padded original."""
+    root = str(tmp_path)
+    _write_cifar(root)
+    plain = RealClassification("Cifar10", root, torch.device("cpu"), augment=False)
+    aug = RealClassification("Cifar10", root, torch.device("cpu"), augment=True)
+    aug2 = RealClassification("Cifar10", root, torch.device("cpu"), augment=True)
+    xa, ya = aug.get_batch(0, 16)
+    xb, yb = aug2.get_batch(0, 16)
+    xp, yp = plain.get_batch(0, 16)
+    assert torch.equal(xa, xb) and torch.equal(ya, yb), "augmentation not deterministic"
+    assert torch.equal(ya, yp), "labels must be untouched"
+    assert xa.shape == xp.shape
+    assert not torch.equal(xa, xp), "augmentation was a no-op"
+    # brute-force verify sample 0 is one of the 9*9*2 crop/flip variants
+    src = torch.nn.functional.pad(xp[0:1], (4, 4, 4, 4))[0]
+    found = False
+    for dy in range(9):
+        for dx in range(9):
+            crop = src[:, dy : dy + 32, dx : dx + 32]
+            if torch.equal(xa[0], crop) or torch.equal(xa[0], crop.flip(2)):
+                found = True
+    assert found, "augmented sample is not a crop/flip of the original"
+
+
+def test_mnist_augment_is_noop(tmp_path):
+    root = str(tmp_path)
+    _write_mnist(root)
+    a = RealClassification("MNIST", root, torch.device("cpu"), augment=True)
+    b = RealClassification("MNIST", root, torch.device("cpu"), augment=False)
+    xa, _ = a.get_batch(0, 8)
+    xb, _ = b.get_batch(0, 8)
+    assert torch.equal(xa, xb)
