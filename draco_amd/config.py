@@ -42,6 +42,10 @@ class Config:
     vote_rtol: float = -1.0           # relative vote tolerance; -1 = auto (0 on CPU, 1e-3 on GPU --
                                       # MIOpen conv backward is not bitwise-reproducible; see
                                       # parallel/aggregators.VoteAggregator)
+    vote_granularity: str = "row"     # row | segment: tolerance ball over the whole
+                                      # gradient vs per parameter tensor (segment is
+                                      # ~10-1000x tighter against within-tolerance
+                                      # adversaries; see VoteAggregator)
     dtype: str = "bf16"               # bf16|fp32 compute dtype (grads/aggregation always fp32)
     device: str = "auto"              # auto|cuda|cpu
     deterministic: bool = False

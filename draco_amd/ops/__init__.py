@@ -23,6 +23,8 @@ __all__ = [
     "combine_rows",
     "cyclic_project",
     "cyclic_recombine",
+    "segment_absmax",
+    "segment_pair_maxdiff",
     "segment_sqdist",
     "segment_weighted_mean",
     "segment_gram",
@@ -151,6 +153,20 @@ def cyclic_recombine(r_planes, v_re, v_im, out):
         ext.cyclic_recombine(r_planes, v_re.to(r_planes.device), v_im.to(r_planes.device), out)
         return
     fallback.cyclic_recombine(r_planes, v_re, v_im, out)
+
+
+def segment_absmax(x, seg):
+    ext = _native_for(x)
+    if ext is not None:
+        return ext.segment_absmax(x, seg.to(x.device))
+    return fallback.segment_absmax(x, seg)
+
+
+def segment_pair_maxdiff(x, a_idx, b_idx, seg):
+    ext = _native_for(x)
+    if ext is not None:
+        return ext.segment_pair_maxdiff(x, a_idx.to(x.device), b_idx.to(x.device), seg.to(x.device))
+    return fallback.segment_pair_maxdiff(x, a_idx, b_idx, seg)
 
 
 def segment_sqdist(x, z, seg):

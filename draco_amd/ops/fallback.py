@@ -204,6 +204,23 @@ def cyclic_recombine(r_planes: torch.Tensor, v_re: torch.Tensor, v_im: torch.Ten
 
 # --------------------------------------------------------------------- geo-median
 @torch.no_grad()
+def segment_absmax(x: torch.Tensor, seg: torch.Tensor) -> torch.Tensor:
+    """(rows, d), (L+1,) bounds -> (rows, L) per-segment max|x| (segment vote)."""
+    L = seg.numel() - 1
+    out = torch.zeros(x.shape[0], L, dtype=torch.float32, device=x.device)
+    for l in range(L):
+        lo, hi = int(seg[l]), int(seg[l + 1])
+        if hi > lo:
+            out[:, l] = x[:, lo:hi].abs().amax(dim=1)
+    return out
+
+
+def segment_pair_maxdiff(x: torch.Tensor, a_idx: torch.Tensor, b_idx: torch.Tensor,
+                         seg: torch.Tensor) -> torch.Tensor:
+    """(pairs, L) per-segment max|x[a]-x[b]| (segment-granular tolerance vote)."""
+    return segment_absmax(x[a_idx] - x[b_idx], seg)
+
+
 def segment_sqdist(x: torch.Tensor, z: torch.Tensor, seg: torch.Tensor) -> torch.Tensor:
     """Per-(row, segment) partial squared distance ||x[p, seg_l] - z[seg_l]||^2.
 

@@ -218,10 +218,18 @@ class VoteAggregator(Aggregator):
 
     def __init__(self, comm, space, group_size: int, atol: float = 0.0,
                  rtol: float = 0.0, member_rows=None, comm_dtype=torch.float32,
-                 member_mask=None):
+                 member_mask=None, granularity: str = "row"):
         super().__init__(comm, space, comm_dtype)
         self.atol = atol
         self.rtol = rtol
+        # granularity="segment": the tolerance ball is PER PARAMETER TENSOR
+        # (max|a-b| <= atol + rtol*max(segmax_a, segmax_b) must hold on EVERY
+        # segment).  This shrinks a within-tolerance adversary's reach from
+        # rtol*max|g| over the whole gradient to rtol*max|g_seg| per tensor —
+        # orders of magnitude tighter for late/small layers — at the cost of a
+        # slightly larger (still tiny) stats allreduce.
+        assert granularity in ("row", "segment"), granularity
+        self.granularity = granularity
         if member_rows is None:
             # colocated layout: G = world groups; member i of group g is local slot i
             # of rank (g+i)%world = row i*world + (g+i)%world (l-major convention of
@@ -280,8 +288,20 @@ class VoteAggregator(Aggregator):
 
     def aggregate(self, payload: torch.Tensor, step: int) -> torch.Tensor:
         recv = self.exchanged(payload)  # (r*world, shard)
-        maxdiff = ops.pair_maxdiff(recv, self.pairs_a, self.pairs_b)  # (n_pairs,)
-        if self.rtol > 0.0:
+        if self.rtol > 0.0 and self.granularity == "segment":
+            npairs = self.pairs_a.numel()
+            segdiff = ops.segment_pair_maxdiff(recv, self.pairs_a, self.pairs_b,
+                                               self.local_seg)  # (n_pairs, L)
+            segmax = ops.segment_absmax(recv, self.local_seg)  # (rows, L)
+            stats = torch.cat([segdiff.reshape(-1), segmax.reshape(-1)])
+            self.comm.all_reduce(stats, op="max")  # per-segment global maxima
+            segdiff = stats[: segdiff.numel()].view(npairs, -1)
+            segmax = stats[segdiff.numel():].view(recv.shape[0], -1)
+            thresh = self.atol + self.rtol * torch.maximum(segmax[self.pairs_a],
+                                                           segmax[self.pairs_b])
+            eq = (segdiff <= thresh).all(dim=1)
+        elif self.rtol > 0.0:
+            maxdiff = ops.pair_maxdiff(recv, self.pairs_a, self.pairs_b)  # (n_pairs,)
             rowmax = ops.row_absmax(recv)  # (world*r,)
             stats = torch.cat([maxdiff, rowmax])
             self.comm.all_reduce(stats, op="max")  # full-gradient maxima
@@ -289,6 +309,7 @@ class VoteAggregator(Aggregator):
             thresh = self.atol + self.rtol * torch.maximum(rowmax[self.pairs_a], rowmax[self.pairs_b])
             eq = maxdiff <= thresh
         else:
+            maxdiff = ops.pair_maxdiff(recv, self.pairs_a, self.pairs_b)
             self.comm.all_reduce(maxdiff, op="max")
             eq = maxdiff <= self.atol
         # winner selection ON DEVICE (every rank computes the identical winners from

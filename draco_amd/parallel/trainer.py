@@ -253,7 +253,8 @@ class Trainer:
             self.agg = VoteAggregator(self.comm, self.space, group_size=self.r,
                                       atol=cfg.vote_atol, rtol=self.vote_rtol,
                                       member_rows=rows, member_mask=mask,
-                                      comm_dtype=self._comm_dtype())
+                                      comm_dtype=self._comm_dtype(),
+                                      granularity=cfg.vote_granularity)
             self.payload = self.space.alloc_payload(self.L)
         else:  # cyclic
             self.agg = CyclicAggregator(self.comm, self.space, self.code, self.L,
@@ -349,8 +350,26 @@ class Trainer:
         if self._hook_adversary:
             # injection at the send boundary, per chunk — same placement as the
             # reference's per-layer err_simulation inside backward (lenet.py:129-141)
-            ops.inject_(self.payload[row, lo:hi], self.cfg.err_mode, cyclic=False)
+            self._inject_row(row, lo, hi)
         self.agg.start_bucket(self.payload, row, lo, hi)
+
+    def _inject_row(self, row: int, lo: int, hi: int) -> None:
+        if self.cfg.err_mode == "within_tol":
+            # worst-case TOLERANCE-RESPECTING adversary (security-model test): a
+            # perturbation just inside the vote's equality ball — it passes the
+            # vote every step and biases the winner by at most the ball radius.
+            # Uses the adversary's own |g| maxima (<= the pairwise max in the
+            # threshold, so the crafted row always stays inside the ball).
+            g = self.payload[row, lo:hi]
+            rtol = getattr(self, "vote_rtol", 0.0)
+            scale = self.cfg.vote_atol + rtol * g.abs().max()
+            gen = torch.Generator(device=g.device)
+            gen.manual_seed(0xADD ^ (self.step_num * 7919) ^ row)
+            sign = torch.randint(0, 2, g.shape, generator=gen, device=g.device,
+                                 dtype=torch.int8).float() * 2.0 - 1.0
+            g.add_(sign * (0.99 * scale))
+        else:
+            ops.inject_(self.payload[row, lo:hi], self.cfg.err_mode, cyclic=False)
 
     def _begin_bucketed_row(self, row: int, adversary: bool) -> None:
         self._bucket_left = [len(b[2]) for b in self._buckets]
@@ -511,7 +530,7 @@ class Trainer:
                 if st is not None:  # order the default stream after worker l's graph
                     torch.cuda.current_stream().wait_stream(st)
                 if worker_id in adversaries:
-                    ops.inject_(self.payload[l], cfg.err_mode, cyclic=False)
+                    self._inject_row(l, 0, self.space.d_pad)
                 # overlap: this row's all_to_all runs while other backwards compute
                 self.agg.start_row(self.payload, l)
         else:  # cyclic
