@@ -155,3 +155,35 @@ def test_segment_kernels():
     gram_ref = fb.segment_gram(x, seg)
     gram_gpu = ops.segment_gram(x.to(DEV), seg.to(DEV))
     assert torch.allclose(gram_ref, gram_gpu.cpu(), rtol=1e-4, atol=1e-2)
+
+
+def test_segment_absmax_kernels_match_oracle():
+    torch.manual_seed(11)
+    rows, d = 9, 8192
+    seg = torch.tensor([0, 3, 700, 701, 4096, 8000], dtype=torch.int64)
+    x = _r(rows, d, seed=12) * torch.logspace(-3, 3, d)  # wide dynamic range
+    ref = fb.segment_absmax(x, seg)
+    got = ops.segment_absmax(x.to(DEV), seg).cpu()
+    assert torch.allclose(ref, got, atol=0, rtol=0), float((ref - got).abs().max())
+
+    a = torch.tensor([0, 1, 4, 7], dtype=torch.int64)
+    b = torch.tensor([2, 3, 5, 8], dtype=torch.int64)
+    ref = fb.segment_pair_maxdiff(x, a, b, seg)
+    got = ops.segment_pair_maxdiff(x.to(DEV), a, b, seg).cpu()
+    assert torch.allclose(ref, got, atol=1e-6), float((ref - got).abs().max())
+
+
+def test_guarded_sgd_skip_is_noop():
+    d = 2048
+    p = _r(d, seed=20).to(DEV)
+    g = _r(d, seed=21).to(DEV)
+    buf = torch.zeros(d, device=DEV)
+    p0 = p.clone()
+    bad = torch.tensor(False, device=DEV)
+    ops.fused_sgd_step(p, g, buf, lr=0.1, momentum=0.9, dampening=0.0,
+                       weight_decay=0.0, nesterov=False, first_step=True, guard=bad)
+    assert torch.equal(p, p0) and torch.equal(buf, torch.zeros_like(buf))
+    ok = torch.tensor(True, device=DEV)
+    ops.fused_sgd_step(p, g, buf, lr=0.1, momentum=0.9, dampening=0.0,
+                       weight_decay=0.0, nesterov=False, first_step=True, guard=ok)
+    assert not torch.equal(p, p0), "guarded-true update must apply"
