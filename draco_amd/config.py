@@ -64,6 +64,8 @@ class Config:
     compile: bool = False         # torch.compile the model before hipGraph capture
     data_root: str = ""           # directory with real MNIST/CIFAR files (IDX/pickle);
                                   # empty or missing files -> deterministic synthetic data
+    synthetic_task: str = "means" # means | teacher (see data/synthetic.py: teacher is
+                                  # the non-saturating convergence-evidence task)
     log_dir: str = "output/logs/"
 
     def sanity(self):

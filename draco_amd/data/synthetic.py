@@ -20,19 +20,41 @@ from ..models import dataset_shape
 
 
 class SyntheticClassification:
-    """Infinite deterministic class-conditional Gaussian image stream."""
+    """Infinite deterministic synthetic image stream, two tasks:
 
-    def __init__(self, dataset: str, device: torch.device, seed: int = 1234, noise: float = 1.0, dtype=torch.float32):
+    task="means"   class-conditional Gaussians (class-mean + noise).  Separable —
+                   a discrimination demo for adversary experiments; clean training
+                   saturates Prec@1 quickly.
+    task="teacher" labels are argmax of a fixed random 2-layer teacher network on
+                   N(0,1) inputs — a genuinely non-trivial decision boundary the
+                   student must LEARN (clean Prec@1 climbs through ~0.5-0.8 over
+                   hundreds of steps instead of saturating), used for the
+                   convergence-under-attack evidence (tools/convergence.py).
+    """
+
+    def __init__(self, dataset: str, device: torch.device, seed: int = 1234, noise: float = 1.0,
+                 dtype=torch.float32, task: str = "means"):
         c, h, w, classes = dataset_shape(dataset)
         self.shape = (c, h, w)
         self.classes = classes
         self.device = device
         self.noise = noise
         self.dtype = dtype
+        self.task = task
         g = torch.Generator(device="cpu")
         g.manual_seed(seed)
         # fixed class means, modest separation so training has to work for it
         self.means = (torch.randn(classes, c, h, w, generator=g) * 0.7).to(device=device, dtype=dtype)
+        if task == "teacher":
+            din = c * h * w
+            hid = 64
+            # Kaiming-ish scaling keeps the teacher logits O(1); the margin
+            # distribution makes a fraction of samples genuinely hard
+            self.t_w1 = (torch.randn(din, hid, generator=g) / din ** 0.5).to(device, dtype)
+            self.t_b1 = (torch.randn(hid, generator=g) * 0.1).to(device, dtype)
+            self.t_w2 = (torch.randn(hid, classes, generator=g) / hid ** 0.5).to(device, dtype)
+        elif task != "means":
+            raise ValueError(f"unknown synthetic task {task!r}")
 
     @staticmethod
     def _mix(start: int) -> int:
@@ -55,6 +77,12 @@ class SyntheticClassification:
         seed = self._mix(start)
         g = torch.Generator(device=self.device)
         g.manual_seed(seed)
+        if self.task == "teacher":
+            x = torch.randn(batch, *self.shape, generator=g, device=self.device,
+                            dtype=self.dtype)
+            h = torch.relu(x.reshape(batch, -1) @ self.t_w1 + self.t_b1)
+            y = (h @ self.t_w2).argmax(dim=1)
+            return x, y
         y = torch.randint(0, self.classes, (batch,), generator=g, device=self.device)
         x = torch.randn(batch, *self.shape, generator=g, device=self.device,
                         dtype=self.dtype) * self.noise

@@ -224,9 +224,10 @@ class Trainer:
         if dataset_available(self.cfg.dataset, self.cfg.data_root):
             return RealClassification(self.cfg.dataset, self.cfg.data_root, self.device,
                                       train=train, augment=train)
-        # held-out synthetic stream: same class means (the task), disjoint index
-        # space (the 2**40 offset in evaluate(); splitmix64 seeding cannot collide)
-        return SyntheticClassification(self.cfg.dataset, self.device, seed=1234)
+        # held-out synthetic stream: same task function, disjoint index space
+        # (the 2**40 offset in evaluate(); splitmix64 seeding cannot collide)
+        return SyntheticClassification(self.cfg.dataset, self.device, seed=1234,
+                                       task=self.cfg.synthetic_task)
 
     def _baseline_aggregator(self, cfg: Config, num_workers: int):
         if cfg.mode == "normal":

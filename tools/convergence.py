@@ -43,10 +43,12 @@ def main():
     p.add_argument("--dataset", type=str, default="Cifar10")
     p.add_argument("--batch-size", type=int, default=128)
     p.add_argument("--lr", type=float, default=0.02)
+    p.add_argument("--task", type=str, default="means", help="means|teacher synthetic task")
     p.add_argument("--out", type=str, default="profiles/convergence")
     args = p.parse_args()
 
     base = dict(network=args.network, dataset=args.dataset, batch_size=args.batch_size,
+                synthetic_task=args.task,
                 lr=args.lr, momentum=0.5, err_mode="rev_grad", max_steps=args.steps + 10,
                 eval_freq=0, log_dir="", train_dir="/tmp/conv_ckpt")
 
