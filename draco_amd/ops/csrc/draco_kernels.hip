@@ -325,6 +325,7 @@ __global__ void k_seg_absmax(const float *__restrict__ x, const long *__restrict
   int cur = -1;
   float mx = 0.f;
   for (; i < d; i += gstride) {
+    if (i < s_seg[0] || i >= s_seg[L]) continue;  // outside all segments (pad tail)
     int lo = 0, hi = L - 1;
     while (lo < hi) {
       int mid = (lo + hi + 1) >> 1;
@@ -530,6 +531,7 @@ __global__ void k_seg_sqdist(const float *__restrict__ x, const float *__restric
   int cur = -1;
   float acc = 0.f;
   for (; i < d; i += gstride) {
+    if (i < s_seg[0] || i >= s_seg[L]) continue;  // outside all segments (pad tail)
     // binary search: segment l with seg[l] <= i < seg[l+1]
     int lo = 0, hi = L - 1;
     while (lo < hi) {

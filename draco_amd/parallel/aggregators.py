@@ -402,6 +402,13 @@ class KrumAggregator(Aggregator):
         super().__init__(comm, space)
         self.num_workers = num_workers
         self.s = s
+        if num_workers > 32:
+            # CPU-side config check: the k_seg_gram LDS tile is sized for P <= 32
+            # (draco_kernels.hip GRAM_CHUNK layout); fail at construction, not at
+            # first kernel launch on the GPU
+            raise ValueError(
+                f"KrumAggregator supports at most 32 workers (got {num_workers}): "
+                "the segment-Gram HIP kernel stages a (P, chunk) LDS tile sized for P <= 32")
         # per-column segment id of the local shard (for device-side winner assembly:
         # out[c] = recv[winner[seg_of(c)], c]); zero-length padding tail maps to 0
         # and is zeroed after the gather

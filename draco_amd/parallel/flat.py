@@ -97,7 +97,15 @@ class FlatSpace:
         for the per-layer comm/compute overlap (post-accumulate-grad hooks launch
         each bucket's exchange while the rest of backward computes).
 
-        Returns [(lo, hi, (param_indices...)), ...] in expected fire order.
+        Range boundaries are snapped UP to 64-float alignment (the slice kernels
+        stream float4): the few head elements a snap excludes belong to the
+        EARLIER-firing neighbour bucket's last param, which is final by the time
+        this bucket ships, so they ride with this (later) bucket.  The first-fired
+        bucket extends to d_pad — the zero padding ships with it, keeping bucketed
+        and whole-row wire traffic identical.
+
+        Returns [(lo, hi, (param_indices...)), ...] in expected fire order;
+        ranges are disjoint and cover [0, d_pad) exactly.
         """
         limit = max(int(bucket_mb * 1024 * 1024 / 4), 1)
         groups, cur, cur_elems = [], [], 0
@@ -110,10 +118,16 @@ class FlatSpace:
         if cur:
             groups.append(cur)
         out = []
+        prev_lo = self.d_pad  # first-fired bucket ends at the padded top
         for group in groups:
             lo = min(self.offsets[i] for i in group)
-            hi = max(self.offsets[i] + self.numels[i] for i in group)
-            out.append((lo, hi, tuple(group)))
+            lo = (lo + ALIGN - 1) // ALIGN * ALIGN if lo > 0 else 0
+            lo = min(lo, prev_lo)
+            out.append((lo, prev_lo, tuple(group)))
+            prev_lo = lo
+        if prev_lo > 0:  # snap of the last group must still cover [0, ..)
+            last_lo, last_hi, idxs = out[-1]
+            out[-1] = (0, last_hi, idxs)
         return out
 
     # ------------------------------------------------------------------ shards
