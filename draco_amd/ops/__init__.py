@@ -147,6 +147,15 @@ def combine_rows(x, rows, w, out):
     fallback.combine_rows(x, rows, w, out)
 
 
+def combine_rows_slice(x, rows, w, out, col_off):
+    """out = sum_i w[i] * x[rows[i], col_off : col_off + len(out)] (bucketed encode)."""
+    ext = _native_for(x)
+    if ext is not None:
+        ext.combine_rows_slice(x, rows.to(x.device), w.to(x.device), out, int(col_off))
+        return
+    fallback.combine_rows_slice(x, rows, w, out, col_off)
+
+
 def cyclic_recombine(r_planes, v_re, v_im, out):
     ext = _native_for(r_planes)
     if ext is not None:

@@ -394,13 +394,25 @@ __global__ void k_combine(const float4 *__restrict__ x, const long *__restrict__
 }
 
 static void launch_combine(torch::Tensor x, torch::Tensor rows, torch::Tensor w,
-                           torch::Tensor out, long stride_elems) {
+                           torch::Tensor out, long stride_elems, long col_off = 0) {
   long d = out.numel();
-  TORCH_CHECK(d % 4 == 0 && stride_elems % 4 == 0, "combine: 16-byte alignment required");
+  TORCH_CHECK(d % 4 == 0 && stride_elems % 4 == 0 && col_off % 4 == 0,
+              "combine: 16-byte alignment required");
   hipLaunchKernelGGL(k_combine, dim3(n_blocks(d / 4, NTHREADS)), dim3(NTHREADS), 0,
-                     cur_stream(), (const float4 *)x.data_ptr<float>(),
+                     cur_stream(), (const float4 *)(x.data_ptr<float>() + col_off),
                      rows.data_ptr<long>(), w.data_ptr<float>(), (int)rows.numel(),
                      (float4 *)out.data_ptr<float>(), d / 4, stride_elems / 4);
+}
+
+// Bucketed cyclic encode (per-layer overlap): combine only the [col_off,
+// col_off + out.numel()) column range of the source rows into `out` (the matching
+// slice of the encoded plane).  Same arithmetic per element as the full-row
+// combine, so bucketed and whole-row encodes are bit-identical.
+void combine_rows_slice(torch::Tensor x, torch::Tensor rows, torch::Tensor w,
+                        torch::Tensor out, long col_off) {
+  CHECK_IN(x); CHECK_IN(rows); CHECK_IN(w); CHECK_IN(out);
+  TORCH_CHECK(col_off + out.numel() <= x.size(1), "combine slice out of range");
+  launch_combine(x, rows, w, out, x.size(1), col_off);
 }
 
 void mean_rows(torch::Tensor x, torch::Tensor idx, torch::Tensor out) {
@@ -665,6 +677,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cyclic_encode", &cyclic_encode);
   m.def("cyclic_project", &cyclic_project);
   m.def("combine_rows", &combine_rows);
+  m.def("combine_rows_slice", &combine_rows_slice);
   m.def("cyclic_recombine", &cyclic_recombine);
   m.def("segment_absmax", &segment_absmax);
   m.def("segment_pair_maxdiff", &segment_pair_maxdiff);

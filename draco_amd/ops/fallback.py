@@ -191,6 +191,17 @@ def combine_rows(x: torch.Tensor, rows: torch.Tensor, w: torch.Tensor, out: torc
 
 
 @torch.no_grad()
+def combine_rows_slice(x: torch.Tensor, rows: torch.Tensor, w: torch.Tensor,
+                       out: torch.Tensor, col_off: int) -> None:
+    """Column-range variant for the bucketed cyclic encode: identical per-element
+    arithmetic to combine_rows restricted to [col_off, col_off + len(out))."""
+    n = out.numel()
+    if n == 0:
+        return
+    out.copy_((w @ x[rows][:, col_off : col_off + n]).reshape(-1))
+
+
+@torch.no_grad()
 def cyclic_recombine(r_planes: torch.Tensor, v_re: torch.Tensor, v_im: torch.Tensor, out: torch.Tensor) -> None:
     """out = Re( v @ R ) over the local shard.
 

@@ -175,7 +175,7 @@ class Trainer:
         # the reference's signature per-layer interleaving (lenet.py:114-218,
         # resnet_split.py:431-623), rebuilt as post-accumulate-grad hooks.
         self.use_buckets = (cfg.bucket_mb > 0 and not self.use_graphs
-                            and approach in ("baseline", "maj_vote")
+                            and approach in ("baseline", "maj_vote", "cyclic")
                             and self.comm.distributed
                             and hasattr(self.agg, "start_bucket"))
         if self.use_buckets:
@@ -346,6 +346,9 @@ class Trainer:
         return hook
 
     def _ship_bucket(self, bi: int) -> None:
+        if self._hook_row == "cyclic":
+            self._ship_bucket_cyclic(bi)
+            return
         lo, hi, _ = self._buckets[bi]
         row = self._hook_row
         if self._hook_adversary:
@@ -353,6 +356,54 @@ class Trainer:
             # reference's per-layer err_simulation inside backward (lenet.py:129-141)
             self._inject_row(row, lo, hi)
         self.agg.start_bucket(self.payload, row, lo, hi)
+
+    # Cyclic per-layer overlap: the encode needs ALL of a rank's sub-batch
+    # gradients, so hooks are armed only during the LAST local sub-batch's
+    # backward — when bucket b of that backward lands, bucket b of every earlier
+    # sub-batch is long final, so the bucket's slice of each logical worker's
+    # encoded planes is computed, injected, and shipped while the rest of the
+    # last backward still runs.
+    def _ship_bucket_cyclic(self, bi: int) -> None:
+        lo, hi, _ = self._buckets[bi]
+        n = hi - lo
+        for l in range(self.L):
+            w_global = l * self.world + self.rank
+            enc = self.payload[2 * l : 2 * l + 2]
+            ops.combine_rows_slice(self.scratch, self._enc_rows[l], self._w_re[l],
+                                   enc[0, lo:hi], lo)
+            ops.combine_rows_slice(self.scratch, self._enc_rows[l], self._w_im[l],
+                                   enc[1, lo:hi], lo)
+            if w_global in self._cyc_advs:
+                self._inject_encoded_slice(enc, lo, hi, self.cfg.err_mode)
+            self.agg.start_bucket(self.payload, 2 * l, lo, hi)
+            self.agg.start_bucket(self.payload, 2 * l + 1, lo, hi)
+
+    def _begin_cyclic_buckets(self, adversaries) -> None:
+        self._bucket_left = [len(b[2]) for b in self._buckets]
+        self._cyc_advs = adversaries
+        self._hook_row = "cyclic"
+
+    def _end_cyclic_buckets(self) -> None:
+        for bi, left in enumerate(self._bucket_left):
+            if left > 0:
+                self._ship_bucket_cyclic(bi)
+        for l in range(self.L):
+            self.agg.mark_row_started(2 * l)
+            self.agg.mark_row_started(2 * l + 1)
+        self._hook_row = None
+
+    def _inject_encoded_slice(self, enc: torch.Tensor, lo: int, hi: int, mode: str) -> None:
+        s = enc[:, lo:hi]
+        if mode == "rev_grad":
+            s.add_(s, alpha=ops.fallback.ADVERSARY_)
+        elif mode == "constant":
+            enc[0, lo:hi].add_(ops.fallback.ADVERSARY_)
+        elif mode in ("random", "none", ""):
+            pass
+        elif mode == "gauss":
+            s.add_(torch.randn_like(s) * s.abs().mean().clamp(min=1e-12) * 100.0)
+        else:
+            raise ValueError(mode)
 
     def _inject_row(self, row: int, lo: int, hi: int) -> None:
         if self.cfg.err_mode == "within_tol":
@@ -537,25 +588,34 @@ class Trainer:
         else:  # cyclic
             # phase 1: every DISTINCT local sub-batch fwd/bwd replays concurrently
             streams = []
+            last = len(self._local_subs) - 1
             for i, j in enumerate(self._local_subs):
                 x, y = self.data.sub_batch(j, step)
+                if self.use_buckets and i == last:
+                    # per-layer overlap: bucketed encode+exchange fires from the
+                    # last sub-batch's backward hooks (see _ship_bucket_cyclic)
+                    self._begin_cyclic_buckets(adversaries)
+                    losses.append(self._forward_backward(x, y, self.scratch[i]))
+                    self._end_cyclic_buckets()
+                    break
                 st = self._worker_stream(i)
                 losses.append(self._run_fwd_bwd(("sub", i), self.scratch[i], x, y, stream=st))
                 streams.append(st)
             for st in streams:
                 if st is not None:
                     torch.cuda.current_stream().wait_stream(st)
-            # phase 2: per logical worker, encode its band (gathered rows), inject,
-            # start the exchange
-            for l in range(self.L):
-                w_global = l * self.world + self.rank
-                enc = self.payload[2 * l : 2 * l + 2]
-                ops.combine_rows(self.scratch, self._enc_rows[l], self._w_re[l], enc[0])
-                ops.combine_rows(self.scratch, self._enc_rows[l], self._w_im[l], enc[1])
-                if w_global in adversaries:
-                    self._inject_encoded(enc, cfg.err_mode)
-                self.agg.start_row(self.payload, 2 * l)
-                self.agg.start_row(self.payload, 2 * l + 1)
+            if not self.use_buckets:
+                # phase 2: per logical worker, encode its band (gathered rows),
+                # inject, start the exchange
+                for l in range(self.L):
+                    w_global = l * self.world + self.rank
+                    enc = self.payload[2 * l : 2 * l + 2]
+                    ops.combine_rows(self.scratch, self._enc_rows[l], self._w_re[l], enc[0])
+                    ops.combine_rows(self.scratch, self._enc_rows[l], self._w_im[l], enc[1])
+                    if w_global in adversaries:
+                        self._inject_encoded(enc, cfg.err_mode)
+                    self.agg.start_row(self.payload, 2 * l)
+                    self.agg.start_row(self.payload, 2 * l + 1)
 
         t_comp = time.perf_counter()
         ev_comp = self._event()

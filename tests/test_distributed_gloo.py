@@ -348,6 +348,13 @@ def test_bucketed_baseline_mean_equal():
     run_dist(_bucket_worker, 2, "baseline", "normal", dict(worker_fail=0))
 
 
+def test_bucketed_cyclic_bitwise_equal():
+    # bucketed encode is the identical per-element arithmetic on column slices,
+    # and the bucket all_to_all moves the same bytes -> bit-for-bit params
+    run_dist(_bucket_worker, 2, "cyclic", "cyclic",
+             dict(worker_fail=1, workers_per_rank=2))
+
+
 def _checkpoint_dist_worker(rank, world, tmpdir):
     """Regression (round-1 advisor, high): rank-0-only save() issued a collective
     sync_buffers that other ranks never matched -> deadlock at the first checkpoint
