@@ -4,6 +4,8 @@ import os
 import pickle
 import struct
 
+import pytest
+
 import numpy as np
 import torch
 
@@ -140,3 +142,27 @@ def test_mnist_augment_is_noop(tmp_path):
     xa, _ = a.get_batch(0, 8)
     xb, _ = b.get_batch(0, 8)
     assert torch.equal(xa, xb)
+
+
+@pytest.mark.gpu
+def test_real_data_training_on_gpu(tmp_path):
+    """Real-file lane on the GPU: parsers + device-side augmentation + trainer."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+    from draco_amd.data.real import RealClassification
+
+    root = str(tmp_path)
+    _write_cifar(root)
+    cfg = Config(network="ResNet18", dataset="Cifar10", batch_size=32, device="cuda",
+                 lr=0.02, approach="maj_vote", mode="maj_vote", group_size=3,
+                 worker_fail=1, data_root=root, max_steps=20, eval_freq=0, log_dir="",
+                 train_dir=str(tmp_path / "ck"))
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    assert isinstance(t.data.data, RealClassification)
+    assert t.data.data.augment
+    for _ in range(3):
+        rec = t.train_step()
+    assert np.isfinite(rec["loss"])
+    assert t.skipped_updates == 0
+    t.close()
