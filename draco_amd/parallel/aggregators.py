@@ -208,10 +208,14 @@ class VoteAggregator(Aggregator):
     MIOpen conv backward is not bitwise-reproducible (measured: enabling
     cudnn.deterministic neither fixes it nor is affordable — 15x on conv backward), so
     honest replicas differ by fp-reorder noise and the DEFAULT GPU rule is the
-    tolerance vote.  This is a deliberate, sound relaxation of the adversary model: an
-    adversary constrained to the tolerance ball of an honest gradient can shift the
-    aggregate by at most tol/G — indistinguishable from fp noise — while anything
-    outside the ball still loses the vote.
+    tolerance vote.  This is a deliberate, MEASURED relaxation of the adversary
+    model (tests/test_vote_security.py): a within-ball adversary that wins its
+    group's tie-break shifts that group's contribution by at most the ball radius
+    (atol + rtol*max|g|) — so the aggregate over G groups by at most s/G of it —
+    and training under a persistent within-ball attack still tracks the clean
+    curve; anything outside the ball loses the vote.  granularity="segment"
+    tightens the ball to per parameter tensor (atol + rtol*max|g_seg| on EVERY
+    segment; honest noise margin measured in profiles/segnoise_r02.txt).
     """
 
     name = "maj_vote"
