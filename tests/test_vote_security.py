@@ -112,13 +112,14 @@ def test_segment_granularity_admits_honest_noise():
     assert float((out - g).abs().max()) < 1e-3
 
 
-def _within_tol_train_worker(rank, world, err_mode):
+def _within_tol_train_worker(rank, world, err_mode, granularity="row"):
     from draco_amd.config import Config
     from draco_amd.parallel.trainer import Trainer
 
     cfg = Config(network="FC", dataset="MNIST", batch_size=8, device="cpu", lr=0.05,
                  approach="maj_vote", mode="maj_vote", group_size=3, worker_fail=1,
-                 err_mode=err_mode, vote_rtol=0.1, max_steps=100, eval_freq=0,
+                 err_mode=err_mode, vote_rtol=0.1, vote_granularity=granularity,
+                 max_steps=100, eval_freq=0,
                  log_dir="", train_dir="/tmp/draco_wt", bucket_mb=0)
     t = Trainer(cfg)
     t.logger.stdout_every = 0
@@ -137,3 +138,14 @@ def test_training_tracks_clean_under_within_tol_attack():
     la, lc = attacked[0][0], clean[0][0]
     assert la[-1] < la[0], "attacked run failed to converge at all"
     assert la[-1] < lc[-1] + 0.3, f"attacked {la[-1]:.3f} vs clean {lc[-1]:.3f}"
+
+
+def test_segment_vote_neutralizes_within_tol_attack_e2e():
+    """Under granularity='segment' the row-ball adversary violates small tensors'
+    per-segment balls and loses every vote: the attacked trajectory should hug the
+    clean one much tighter than under the row vote."""
+    attacked = run_dist(_within_tol_train_worker, 3, "within_tol", "segment")
+    clean = run_dist(_within_tol_train_worker, 3, "none", "segment")
+    la, lc = attacked[0][0], clean[0][0]
+    assert attacked[0][1] == attacked[1][1] == attacked[2][1]
+    assert abs(la[-1] - lc[-1]) < 0.1, f"attacked {la[-1]:.3f} vs clean {lc[-1]:.3f}"
