@@ -130,25 +130,6 @@ def main():
     final_loss = t.train_step()["loss"]
     ms_per_step = elapsed / args.steps * 1000.0
 
-    # ---- plain-DP reference arm: same model/batch/world, approach=baseline mode=
-    # normal, NO adversary — quantifies the coding overhead on the same box.
-    # vs_baseline = coded images/sec ÷ plain-DP images/sec (the reference repo
-    # publishes no numbers, BASELINE.md, so the plain arm is the denominator).
-    vs_baseline = None
-    baseline_ips = None
-    if args.approach != "baseline":
-        import dataclasses
-
-        bcfg = dataclasses.replace(
-            cfg, approach="baseline", mode="normal", worker_fail=0, err_mode="none",
-            train_dir="gpurun_out/bench_ckpt_base")
-        tb = Trainer(bcfg)
-        tb.logger.stdout_every = 0
-        tb.logger.close()
-        bsteps = max(min(args.steps, 10), 1)
-        belapsed = timed_region(tb, bsteps, min(args.warmup, 3))
-        baseline_ips = world * args.batch_size * bsteps / belapsed
-        tb.close()
     # DISTINCT images per step (redundant compute is the price of the code and is
     # not counted): maj_vote -> G*B (G=world groups); cyclic -> n*B global batch;
     # baseline -> world*B
@@ -163,6 +144,29 @@ def main():
     else:
         distinct_per_step = world * args.batch_size
         processed = distinct_per_step
+
+    # ---- plain-DP reference arm: same model/world, approach=baseline mode=normal,
+    # NO adversary, processing the SAME distinct images per step as the coded arm
+    # (cyclic trains an n*B global batch, so the fair plain-DP rank batch is
+    # distinct/world) — vs_baseline = coded ÷ plain throughput at equal work.
+    # (The reference repo publishes no numbers, BASELINE.md, so the plain arm is
+    # the denominator.)
+    vs_baseline = None
+    baseline_ips = None
+    if args.approach != "baseline":
+        import dataclasses
+
+        bcfg = dataclasses.replace(
+            cfg, approach="baseline", mode="normal", worker_fail=0, err_mode="none",
+            batch_size=max(distinct_per_step // world, 1),
+            train_dir="gpurun_out/bench_ckpt_base")
+        tb = Trainer(bcfg)
+        tb.logger.stdout_every = 0
+        tb.logger.close()
+        bsteps = max(min(args.steps, 10), 1)
+        belapsed = timed_region(tb, bsteps, min(args.warmup, 3))
+        baseline_ips = distinct_per_step * bsteps / belapsed
+        tb.close()
     images_per_sec = distinct_per_step * args.steps / elapsed
     if baseline_ips is not None:
         vs_baseline = round(images_per_sec / baseline_ips, 4)
