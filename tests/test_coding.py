@@ -93,3 +93,28 @@ def test_cyclic_rejects_bad_params():
         build_cyclic_code(3, 1)  # n < 2s+2
     with pytest.raises(ValueError):
         build_cyclic_code(8, 0)
+
+
+def test_colocated_member_rows_full_and_survivor():
+    from draco_amd.coding import colocated_member_rows
+
+    G, r, world = 4, 3, 4
+    rows, mask = colocated_member_rows(G, r, world, alive=list(range(world)))
+    assert mask.all()
+    # full world: member i of group g at row i*world + (g+i)%world
+    for g in range(G):
+        for i in range(r):
+            assert rows[g, i] == i * world + (g + i) % world
+    # rank 2 dead: its hosted members are forfeited, others re-indexed over [0,1,3]
+    alive = [0, 1, 3]
+    rows2, mask2 = colocated_member_rows(G, r, world, alive)
+    pos = {0: 0, 1: 1, 3: 2}
+    for g in range(G):
+        for i in range(r):
+            h = (g + i) % world
+            if h == 2:
+                assert not mask2[g, i]
+            else:
+                assert mask2[g, i] and rows2[g, i] == i * 3 + pos[h]
+    # every group loses at most one member with one dead rank
+    assert (mask2.sum(axis=1) >= r - 1).all()

@@ -128,3 +128,31 @@ def test_checkpoint_cross_layout_resume(tmp_path):
     for o, n, shape in zip(s1.offsets, s1.numels, s1.shapes):
         assert torch.allclose(s1._view(s1.flat_param, o, n, shape),
                               s2._view(s2.flat_param, o, n, shape), atol=1e-6)
+
+
+def test_build_buckets_invariants():
+    """Bucket ranges: disjoint, aligned, reverse-order, exactly covering [0, d_pad)."""
+    import torch.nn as nn
+    from draco_amd.parallel.flat import ALIGN, FlatSpace
+
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Linear(333, 77), nn.Linear(77, 55), nn.Linear(55, 11),
+                          nn.Linear(11, 3))
+    for world in (1, 2, 3):
+        space = FlatSpace(model, world, torch.device("cpu"))
+        for mb in (0.001, 0.01, 10.0):
+            buckets = space.build_buckets(mb)
+            # coverage: ranges tile [0, d_pad) in reverse order
+            expect_hi = space.d_pad
+            seen = set()
+            for lo, hi, idxs in buckets:
+                assert hi == expect_hi, "ranges must tile top-down"
+                assert lo < hi
+                assert lo % ALIGN == 0 or lo == 0
+                seen.update(idxs)
+                expect_hi = lo
+            assert expect_hi == 0
+            assert seen == set(range(len(space.params)))
+            # groups are consecutive reverse runs of param indices
+            flat_idx = [i for _, _, idxs in buckets for i in idxs]
+            assert flat_idx == list(reversed(range(len(space.params))))
