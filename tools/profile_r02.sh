@@ -11,12 +11,12 @@ mkdir -p gpurun_out/p2
 timeout 420 rocprofv3 --output-format csv --kernel-trace --stats -d gpurun_out/p2 -o steady_r02 -- \
   python bench.py --steps 40 --warmup 10 > gpurun_out/prof_a.log 2>&1
 echo "P1=$?"
-rm -f gpurun_out/p2/*results.db gpurun_out/p2/*.db
+rm -f gpurun_out/p2/*results.db gpurun_out/p2/*.db gpurun_out/p2/*_kernel_trace.csv
 timeout 420 rocprofv3 --output-format csv --kernel-trace --stats -d gpurun_out/p2 -o cyclic_r02 -- \
   python bench.py --steps 40 --warmup 10 --approach cyclic --worker-fail 1 \
   > gpurun_out/prof_b.log 2>&1
 echo "P2=$?"
-rm -f gpurun_out/p2/*results.db gpurun_out/p2/*.db
+rm -f gpurun_out/p2/*results.db gpurun_out/p2/*.db gpurun_out/p2/*_kernel_trace.csv
 timeout 420 rocprofv3 --output-format csv --pmc GRBM_GUI_ACTIVE,SQ_BUSY_CYCLES,SQ_VALU_MFMA_BUSY_CYCLES,SQ_WAVE_CYCLES \
   -d gpurun_out/p2 -o pmc_r02 -- python bench.py --steps 10 --warmup 4 \
   > gpurun_out/prof_c.log 2>&1
