@@ -18,14 +18,14 @@ from draco_amd.data.real import (
 )
 
 
-def _write_mnist(root, n=64):
-    rng = np.random.default_rng(0)
+def _write_mnist(root, n=64, prefix="train", seed=0):
+    rng = np.random.default_rng(seed)
     imgs = rng.integers(0, 256, size=(n, 28, 28), dtype=np.uint8)
     labels = rng.integers(0, 10, size=n, dtype=np.uint8)
-    with open(os.path.join(root, "train-images-idx3-ubyte"), "wb") as f:
+    with open(os.path.join(root, f"{prefix}-images-idx3-ubyte"), "wb") as f:
         f.write(struct.pack(">IIII", 2051, n, 28, 28))
         f.write(imgs.tobytes())
-    with gzip.open(os.path.join(root, "train-labels-idx1-ubyte.gz"), "wb") as f:
+    with gzip.open(os.path.join(root, f"{prefix}-labels-idx1-ubyte.gz"), "wb") as f:
         f.write(struct.pack(">II", 2049, n))
         f.write(labels.tobytes())
     return imgs, labels
@@ -165,4 +165,31 @@ def test_real_data_training_on_gpu(tmp_path):
         rec = t.train_step()
     assert np.isfinite(rec["loss"])
     assert t.skipped_updates == 0
+    t.close()
+
+
+def test_evaluate_uses_test_split(tmp_path):
+    """Regression (round-1 advisor, medium): evaluate() must score the HELD-OUT
+    split, not training data."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    root = str(tmp_path)
+    _write_mnist(root, prefix="train", seed=0)
+    test_imgs, test_labels = _write_mnist(root, prefix="t10k", seed=99)
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                 approach="baseline", mode="normal", worker_fail=0, data_root=root,
+                 test_batch_size=8, max_steps=20, eval_freq=0, log_dir="",
+                 train_dir=str(tmp_path / "ck"))
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    held = t._dataset(train=False)
+    x, y = held.get_batch(0, 8)
+    # the held-out source must serve t10k content (seed 99), not train content
+    norm = (test_imgs.astype(np.float32) / 255.0 - 0.1307) / 0.3081
+    pool = torch.from_numpy(norm).unsqueeze(1)
+    match = any(torch.allclose(x[0], pool[i], atol=1e-5) for i in range(len(pool)))
+    assert match, "evaluate() source does not serve the test split"
+    m = t.evaluate(n_batches=2)
+    assert 0.0 <= m["prec1"] <= 1.0
     t.close()
