@@ -193,3 +193,19 @@ def test_evaluate_uses_test_split(tmp_path):
     m = t.evaluate(n_batches=2)
     assert 0.0 <= m["prec1"] <= 1.0
     t.close()
+
+
+def test_data_prepare_tool(tmp_path):
+    """Reference-parity dataset prep entry point (data_prepare.py): synthesize mode
+    writes real-format files that the loaders then accept."""
+    from draco_amd.data.prepare import main
+
+    root = str(tmp_path / "d")
+    assert main(["--root", root, "--synthesize"]) == 0
+    assert dataset_available("MNIST", root) and dataset_available("Cifar10", root)
+    x, y = load_mnist_idx(root, train=False)
+    assert x.shape[0] == 1024
+    x2, y2 = load_cifar10(root, train=True)
+    assert x2.shape[0] == 5 * 1024
+    # verify-only mode on an empty dir reports missing
+    assert main(["--root", str(tmp_path / "empty")]) != 0
