@@ -309,36 +309,56 @@ torch::Tensor row_absmax(torch::Tensor x) {
 // rtol*rowmax over the whole gradient (~10x tighter for late small layers).
 // Same layout as k_seg_sqdist: segment bounds in LDS, binary search per element,
 // flush-on-segment-change with a float atomic max.
+// The grid-stride loop makes a thread's CONSECUTIVE elements gstride apart, i.e.
+// almost always in DIFFERENT segments — so flush-on-segment-change degenerates to
+// one GLOBAL atomic per element (measured 4-7 GB/s).  Instead each block
+// accumulates into an LDS per-segment array (LDS atomics: conflict-serialised but
+// ~3 orders cheaper than L2) and flushes ONCE per (block, segment).
 template <bool PAIR>
 __global__ void k_seg_absmax(const float *__restrict__ x, const long *__restrict__ ai,
                              const long *__restrict__ bi, const long *__restrict__ seg,
                              int L, float *__restrict__ out /* (rows, L) */, long d,
                              long stride) {
-  extern __shared__ long s_seg[];
+  extern __shared__ char smem[];
+  long *s_seg = (long *)smem;
+  int *s_max = (int *)(smem + (L + 1) * sizeof(long));  // float bits (non-negative)
   for (int i = threadIdx.x; i <= L; i += blockDim.x) s_seg[i] = seg[i];
+  for (int i = threadIdx.x; i < L; i += blockDim.x) s_max[i] = 0;
   __syncthreads();
   int r = blockIdx.y;
   const float *ra = x + (PAIR ? ai[r] : (long)r) * stride;
   const float *rb = PAIR ? (x + bi[r] * stride) : nullptr;
-  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
   long gstride = gridDim.x * (long)blockDim.x;
-  int cur = -1;
-  float mx = 0.f;
-  for (; i < d; i += gstride) {
-    if (i < s_seg[0] || i >= s_seg[L]) continue;  // outside all segments (pad tail)
-    int lo = 0, hi = L - 1;
-    while (lo < hi) {
-      int mid = (lo + hi + 1) >> 1;
-      if (s_seg[mid] <= i) lo = mid; else hi = mid - 1;
+  // base-indexed loop: the trip condition is uniform per block, so every lane is
+  // active at each __shfl/__all (a wave whose 64 lanes share a segment — the
+  // common case — pre-reduces in registers and issues ONE LDS atomic)
+  for (long base = blockIdx.x * (long)blockDim.x; base < d; base += gstride) {
+    long i = base + threadIdx.x;
+    bool valid = (i < d) && i >= s_seg[0] && i < s_seg[L];
+    int lo = 0;
+    float v = 0.f;
+    if (valid) {
+      int hi = L - 1;
+      while (lo < hi) {
+        int mid = (lo + hi + 1) >> 1;
+        if (s_seg[mid] <= i) lo = mid; else hi = mid - 1;
+      }
+      v = fabsf(PAIR ? (ra[i] - rb[i]) : ra[i]);
     }
-    float v = PAIR ? (ra[i] - rb[i]) : ra[i];
-    if (lo != cur) {
-      if (cur >= 0 && mx > 0.f) atomic_max_f32(&out[r * L + cur], mx);
-      cur = lo; mx = 0.f;
+    int lo0 = __shfl(lo, 0, WAVE);
+    if (__all(valid && lo == lo0)) {
+#pragma unroll
+      for (int off = WAVE / 2; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off, WAVE));
+      if ((threadIdx.x & (WAVE - 1)) == 0) atomicMax(&s_max[lo0], __float_as_int(v));
+    } else if (valid) {
+      atomicMax(&s_max[lo], __float_as_int(v));
     }
-    mx = fmaxf(mx, fabsf(v));
   }
-  if (cur >= 0 && mx > 0.f) atomic_max_f32(&out[r * L + cur], mx);
+  __syncthreads();
+  for (int l = threadIdx.x; l < L; l += blockDim.x) {
+    float m = __int_as_float(s_max[l]);
+    if (m > 0.f) atomic_max_f32(&out[r * L + l], m);
+  }
 }
 
 torch::Tensor segment_absmax(torch::Tensor x, torch::Tensor seg) {
@@ -348,7 +368,8 @@ torch::Tensor segment_absmax(torch::Tensor x, torch::Tensor seg) {
   auto out = torch::zeros({m, L}, torch::dtype(torch::kFloat32).device(x.device()));
   if (d == 0 || m == 0) return out;
   dim3 grid(n_blocks(d / 8, NTHREADS), (unsigned)m);
-  hipLaunchKernelGGL((k_seg_absmax<false>), grid, dim3(NTHREADS), (L + 1) * sizeof(long),
+  size_t smem = (L + 1) * sizeof(long) + L * sizeof(int);
+  hipLaunchKernelGGL((k_seg_absmax<false>), grid, dim3(NTHREADS), smem,
                      cur_stream(), x.data_ptr<float>(), nullptr, nullptr,
                      seg.data_ptr<long>(), L, out.data_ptr<float>(), d, d);
   return out;
@@ -362,7 +383,8 @@ torch::Tensor segment_pair_maxdiff(torch::Tensor x, torch::Tensor a_idx,
   auto out = torch::zeros({k, L}, torch::dtype(torch::kFloat32).device(x.device()));
   if (d == 0 || k == 0) return out;
   dim3 grid(n_blocks(d / 8, NTHREADS), (unsigned)k);
-  hipLaunchKernelGGL((k_seg_absmax<true>), grid, dim3(NTHREADS), (L + 1) * sizeof(long),
+  size_t smem = (L + 1) * sizeof(long) + L * sizeof(int);
+  hipLaunchKernelGGL((k_seg_absmax<true>), grid, dim3(NTHREADS), smem,
                      cur_stream(), x.data_ptr<float>(), a_idx.data_ptr<long>(),
                      b_idx.data_ptr<long>(), seg.data_ptr<long>(), L,
                      out.data_ptr<float>(), d, d);
@@ -530,34 +552,49 @@ void combine_rows(torch::Tensor x, torch::Tensor rows, torch::Tensor w, torch::T
 // K6 (baseline_master.py:271-276 / hdmedians): per-(worker, layer-segment) partial
 // squared distances for the sharded Weiszfeld iteration.  Segment bounds live in LDS;
 // each thread keeps a running (segment, acc) and flushes on change via atomicAdd.
+// Same LDS-accumulate pattern as k_seg_absmax (see comment there): grid-stride
+// means per-element segment changes, so per-thread flush-on-change would cost one
+// GLOBAL atomic per element.  Blocks accumulate per-segment partial sums in LDS
+// and flush once per (block, segment).
 __global__ void k_seg_sqdist(const float *__restrict__ x, const float *__restrict__ z,
                              const long *__restrict__ seg, int L,
                              float *__restrict__ out /* (P, L) */, long d, long stride) {
-  extern __shared__ long s_seg[];
+  extern __shared__ char smem[];
+  long *s_seg = (long *)smem;
+  float *s_acc = (float *)(smem + (L + 1) * sizeof(long));
   for (int i = threadIdx.x; i <= L; i += blockDim.x) s_seg[i] = seg[i];
+  for (int i = threadIdx.x; i < L; i += blockDim.x) s_acc[i] = 0.f;
   __syncthreads();
   int p = blockIdx.y;
   const float *row = x + p * stride;
-  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
   long gstride = gridDim.x * (long)blockDim.x;
-  int cur = -1;
-  float acc = 0.f;
-  for (; i < d; i += gstride) {
-    if (i < s_seg[0] || i >= s_seg[L]) continue;  // outside all segments (pad tail)
-    // binary search: segment l with seg[l] <= i < seg[l+1]
-    int lo = 0, hi = L - 1;
-    while (lo < hi) {
-      int mid = (lo + hi + 1) >> 1;
-      if (s_seg[mid] <= i) lo = mid; else hi = mid - 1;
+  for (long base = blockIdx.x * (long)blockDim.x; base < d; base += gstride) {
+    long i = base + threadIdx.x;
+    bool valid = (i < d) && i >= s_seg[0] && i < s_seg[L];
+    int lo = 0;
+    float c = 0.f;
+    if (valid) {
+      // binary search: segment l with seg[l] <= i < seg[l+1]
+      int hi = L - 1;
+      while (lo < hi) {
+        int mid = (lo + hi + 1) >> 1;
+        if (s_seg[mid] <= i) lo = mid; else hi = mid - 1;
+      }
+      float dlt = row[i] - z[i];
+      c = dlt * dlt;
     }
-    float dlt = row[i] - z[i];
-    if (lo != cur) {
-      if (cur >= 0 && acc != 0.f) atomicAdd(&out[p * L + cur], acc);
-      cur = lo; acc = 0.f;
+    int lo0 = __shfl(lo, 0, WAVE);
+    if (__all(valid && lo == lo0)) {  // whole wave in one segment: one LDS atomic
+#pragma unroll
+      for (int off = WAVE / 2; off > 0; off >>= 1) c += __shfl_down(c, off, WAVE);
+      if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(&s_acc[lo0], c);
+    } else if (valid) {
+      atomicAdd(&s_acc[lo], c);
     }
-    acc = fmaf(dlt, dlt, acc);
   }
-  if (cur >= 0 && acc != 0.f) atomicAdd(&out[p * L + cur], acc);
+  __syncthreads();
+  for (int l = threadIdx.x; l < L; l += blockDim.x)
+    if (s_acc[l] != 0.f) atomicAdd(&out[p * L + l], s_acc[l]);
 }
 
 torch::Tensor segment_sqdist(torch::Tensor x, torch::Tensor z, torch::Tensor seg) {
@@ -567,7 +604,8 @@ torch::Tensor segment_sqdist(torch::Tensor x, torch::Tensor z, torch::Tensor seg
   auto out = torch::zeros({P, L}, torch::dtype(torch::kFloat32).device(x.device()));
   if (d == 0) return out;
   dim3 grid(n_blocks(d / 4, NTHREADS), (unsigned)P);
-  hipLaunchKernelGGL(k_seg_sqdist, grid, dim3(NTHREADS), (L + 1) * sizeof(long),
+  size_t smem = (L + 1) * sizeof(long) + L * sizeof(float);
+  hipLaunchKernelGGL(k_seg_sqdist, grid, dim3(NTHREADS), smem,
                      cur_stream(), x.data_ptr<float>(), z.data_ptr<float>(),
                      seg.data_ptr<long>(), L, out.data_ptr<float>(), d, d);
   return out;
@@ -610,11 +648,15 @@ void segment_weighted_mean(torch::Tensor x, torch::Tensor w, torch::Tensor seg,
 // K7 (baseline_master.py:278-296): per-segment Gram matrices for the pairwise
 // distance matrix.  Host pre-splits segments into <=CHUNK-element descriptors; a block
 // stages its chunk (P x CHUNK) through LDS and accumulates all P*P pair dots.
-#define GRAM_CHUNK 512
+#define GRAM_CHUNK 256
+#define GRAM_LD (GRAM_CHUNK + 1)  // +1 float: row stride odd mod 64 banks, so
+                                  // lanes reading consecutive rows hit distinct
+                                  // banks (stride 256/512 put EVERY lane on one
+                                  // bank: measured 64-way conflict, ~50 GB/s)
 __global__ void k_seg_gram(const float *__restrict__ x, const long *__restrict__ desc,
                            /* desc: (nchunks, 3) = (seg, start, len) */
                            int P, float *__restrict__ out /* (L, P, P) */, long stride) {
-  __shared__ float tile[32][GRAM_CHUNK / 4][4];  // [P][chunk] as float4 groups
+  __shared__ float tile[32][GRAM_LD];  // [P][chunk], padded leading dim
   int c = blockIdx.x;
   long segid = desc[c * 3 + 0];
   long start = desc[c * 3 + 1];
@@ -623,7 +665,7 @@ __global__ void k_seg_gram(const float *__restrict__ x, const long *__restrict__
   for (long t = threadIdx.x; t < (long)P * len; t += blockDim.x) {
     int p = (int)(t / len);
     long e = t % len;
-    tile[p][e / 4][e % 4] = x[p * stride + start + e];
+    tile[p][e] = x[p * stride + start + e];
   }
   __syncthreads();
   float *base = out + segid * P * P;
@@ -632,7 +674,7 @@ __global__ void k_seg_gram(const float *__restrict__ x, const long *__restrict__
     if (b < a) continue;  // symmetric: fill upper, mirror below
     float acc = 0.f;
     for (long e = 0; e < len; ++e)
-      acc = fmaf(tile[a][e / 4][e % 4], tile[b][e / 4][e % 4], acc);
+      acc = fmaf(tile[a][e], tile[b][e], acc);
     atomicAdd(&base[a * P + b], acc);
     if (a != b) atomicAdd(&base[b * P + a], acc);
   }
