@@ -668,15 +668,25 @@ __global__ void k_seg_gram(const float *__restrict__ x, const long *__restrict__
     tile[p][e] = x[p * stride + start + e];
   }
   __syncthreads();
+  // one PAIR per wave, lanes parallel over the chunk (a per-thread serial dot
+  // product is a ~256-deep dependent FMA/LDS chain — measured 60 GB/s; the
+  // wave-parallel form is 64 lanes x independent partials + one shuffle reduce)
   float *base = out + segid * P * P;
-  for (int pair = threadIdx.x; pair < P * P; pair += blockDim.x) {
+  int wid = threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int nwaves = blockDim.x / WAVE;
+  for (int pair = wid; pair < P * P; pair += nwaves) {
     int a = pair / P, b = pair % P;
     if (b < a) continue;  // symmetric: fill upper, mirror below
     float acc = 0.f;
-    for (long e = 0; e < len; ++e)
+    for (long e = lane; e < len; e += WAVE)
       acc = fmaf(tile[a][e], tile[b][e], acc);
-    atomicAdd(&base[a * P + b], acc);
-    if (a != b) atomicAdd(&base[b * P + a], acc);
+#pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1) acc += __shfl_down(acc, off, WAVE);
+    if (lane == 0) {
+      atomicAdd(&base[a * P + b], acc);
+      if (a != b) atomicAdd(&base[b * P + a], acc);
+    }
   }
 }
 
