@@ -668,25 +668,24 @@ __global__ void k_seg_gram(const float *__restrict__ x, const long *__restrict__
     tile[p][e] = x[p * stride + start + e];
   }
   __syncthreads();
-  // one PAIR per wave, lanes parallel over the chunk (a per-thread serial dot
-  // product is a ~256-deep dependent FMA/LDS chain — measured 60 GB/s; the
-  // wave-parallel form is 64 lanes x independent partials + one shuffle reduce)
   float *base = out + segid * P * P;
-  int wid = threadIdx.x / WAVE;
-  int lane = threadIdx.x % WAVE;
-  int nwaves = blockDim.x / WAVE;
-  for (int pair = wid; pair < P * P; pair += nwaves) {
+  for (int pair = threadIdx.x; pair < P * P; pair += blockDim.x) {
     int a = pair / P, b = pair % P;
     if (b < a) continue;  // symmetric: fill upper, mirror below
-    float acc = 0.f;
-    for (long e = lane; e < len; e += WAVE)
-      acc = fmaf(tile[a][e], tile[b][e], acc);
-#pragma unroll
-    for (int off = WAVE / 2; off > 0; off >>= 1) acc += __shfl_down(acc, off, WAVE);
-    if (lane == 0) {
-      atomicAdd(&base[a * P + b], acc);
-      if (a != b) atomicAdd(&base[b * P + a], acc);
+    // 4 independent accumulator chains: the naive single-chain form is a
+    // len-deep dependent FMA+LDS sequence (latency-bound, measured ~60 GB/s)
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    long e = 0;
+    for (; e + 3 < len; e += 4) {
+      a0 = fmaf(tile[a][e], tile[b][e], a0);
+      a1 = fmaf(tile[a][e + 1], tile[b][e + 1], a1);
+      a2 = fmaf(tile[a][e + 2], tile[b][e + 2], a2);
+      a3 = fmaf(tile[a][e + 3], tile[b][e + 3], a3);
     }
+    float acc = (a0 + a1) + (a2 + a3);
+    for (; e < len; ++e) acc = fmaf(tile[a][e], tile[b][e], acc);
+    atomicAdd(&base[a * P + b], acc);
+    if (a != b) atomicAdd(&base[b * P + a], acc);
   }
 }
 

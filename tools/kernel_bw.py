@@ -92,5 +92,24 @@ def main():
     report("segment_gram (krum)", ms, B)
 
 
+def gram_compare():
+    """Hand-written k_seg_gram vs the rocBLAS route (62 torch.mm per-segment GEMMs
+    — the brief's 'plain library GEMMs belong on MFMA' case)."""
+    from draco_amd.ops import fallback as fb
+    d = 11_173_952
+    d = (d + 63) // 64 * 64
+    rows = 24
+    x = torch.randn(rows, d, device=DEV)
+    seg = torch.linspace(0, d, 63, dtype=torch.int64).to(DEV)
+    B = 4 * rows * d
+    ms = t_ms(lambda: ops.segment_gram(x, seg), iters=10, warmup=3)
+    report("segment_gram HIP kernel", ms, B)
+    ms = t_ms(lambda: fb.segment_gram(x, seg), iters=10, warmup=3)
+    report("segment_gram rocBLAS (62 mm)", ms, B)
+
+
 if __name__ == "__main__":
+    if os.environ.get("GRAM_ONLY") == "1":
+        gram_compare()
+        raise SystemExit(0)
     main()
