@@ -71,3 +71,40 @@ def test_cyclic_survives_rank_death():
 
 def test_baseline_mean_survives_rank_death():
     _run_failure(3, "baseline", dict(mode="normal", worker_fail=0))
+
+
+def _combined_worker(rank, world, die_rank, die_step):
+    """Erasure AND adversary together: r=5 groups lose <=2 members to the dead
+    rank while a rev_grad adversary keeps striking the survivors — each group
+    still has an honest majority among its alive members, so the vote holds."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                 approach="maj_vote", mode="maj_vote", group_size=5, worker_fail=1,
+                 err_mode="rev_grad", max_steps=100, eval_freq=0, log_dir="",
+                 train_dir="/tmp/draco_fail2", health_timeout=1.5)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    losses = []
+    for i in range(12):
+        if rank == die_rank and i == die_step:
+            time.sleep(0.2)
+            os._exit(0)
+        losses.append(t.train_step()["loss"])
+    h = float(t.space.flat_param.double().sum())
+    deg = t.agg.degenerate_steps
+    t.close()
+    return (losses, h, deg)
+
+
+def test_vote_survives_rank_death_with_active_adversary():
+    world, die_rank = 3, 2
+    res = run_dist(_combined_worker, world, die_rank, 3, timeout=240.0,
+                   expect_missing={die_rank})
+    survivors = [0, 1]
+    assert res[0][1] == res[1][1], "survivor params diverged"
+    for r in survivors:
+        losses = res[r][0]
+        assert losses[-1] < losses[0], "did not keep converging"
+        assert res[r][2] == 0, "vote degenerated under combined erasure+adversary"
