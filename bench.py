@@ -195,6 +195,7 @@ def main():
             "final_loss": round(final_loss, 4) if final_loss == final_loss else None,
             "skipped_updates": t.skipped_updates,
             "vote_degenerate_steps": getattr(t.agg, "degenerate_steps", None),
+            "peak_mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 3) if on_gpu else None,
             "dtype": args.dtype if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
