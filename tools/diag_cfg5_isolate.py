@@ -43,6 +43,12 @@ def run(name, steps, serial=False, **kw):
     return spike
 
 
+def matrix():
+    run("graphs-nchw", 120, channels_last=False)
+    run("graphs-fp32", 120, dtype="fp32")
+    run("graphs-no-adversary", 120, err_mode="none")
+
+
 if __name__ == "__main__":
     if os.environ.get("MATRIX") == "1":
         matrix()
@@ -50,9 +56,3 @@ if __name__ == "__main__":
     run("concurrent-graphs", 120)
     run("serial-graphs", 120, serial=True)
     run("eager", 120, hip_graphs=False)
-
-
-def matrix():
-    run("graphs-nchw", 120, channels_last=False)
-    run("graphs-fp32", 120, dtype="fp32")
-    run("graphs-no-adversary", 120, err_mode="none")
