@@ -14,4 +14,4 @@ run cfg2_rep_r18        # defaults: ResNet18 maj_vote r=3 s=1
 run cfg3_cyc_r18        --approach cyclic --mode cyclic --worker-fail 1
 run cfg4_geomed_vgg11   --approach baseline --mode geometric_median --network VGG11 --worker-fail 2
 run cfg4_cyc_vgg11      --approach cyclic --mode cyclic --network VGG11 --worker-fail 2
-run cfg5_cyc_r50        --approach cyclic --mode cyclic --worker-fail 2 --network ResNet50 --dataset ImageNetSynthetic --batch-size 32
+run cfg5_cyc_r50        --approach cyclic --mode cyclic --worker-fail 2 --network ResNet50 --dataset ImageNetSynthetic --batch-size 32 --channels-last false

@@ -8,9 +8,9 @@
 set -x
 for i in 1 2; do
   timeout 600 python bench.py --steps 200 --warmup 3 --approach cyclic --worker-fail 2 \
-    --network ResNet50 --dataset ImageNetSynthetic --batch-size 32 --compile false \
+    --network ResNet50 --dataset ImageNetSynthetic --batch-size 32 --compile false --channels-last false \
     2>/dev/null | python -c "import json,sys; r=json.load(sys.stdin); print('graphs soak', r['ms_per_step'], 'ms/step loss', r['final_loss'], 'skipped', r['skipped_updates'])"
 done
 timeout 400 python bench.py --steps 10 --warmup 3 --approach cyclic --worker-fail 2 \
-  --network ResNet50 --dataset ImageNetSynthetic --batch-size 32 --compile false --hip-graphs false \
+  --network ResNet50 --dataset ImageNetSynthetic --batch-size 32 --compile false --channels-last false --hip-graphs false \
   2>/dev/null | python -c "import json,sys; r=json.load(sys.stdin); print('eager run', r['ms_per_step'], 'ms/step loss', r['final_loss'], 'skipped', r['skipped_updates'])"
