@@ -148,3 +148,23 @@ def test_gpu_checkpoint_roundtrip(tmp_path):
     # backward is not bitwise-reproducible on MIOpen: trajectories match loosely
     assert np.allclose(ref, resumed, rtol=0.1, atol=0.05), (ref, resumed)
     t2.close()
+
+
+def test_nhwc_graph_guard_for_224_shapes(tmp_path):
+    """bf16+channels_last+hipGraphs corrupts 224x224 conv replays on this stack
+    (KNOWN_ISSUES #cfg5-nhwc): the trainer must force NCHW on that path."""
+    import warnings
+
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="ResNet50", dataset="ImageNetSynthetic", batch_size=4,
+                 approach="cyclic", mode="cyclic", worker_fail=2, device="cuda",
+                 dtype="bf16", channels_last=True, hip_graphs=True,
+                 max_steps=5, eval_freq=0, log_dir="", train_dir=str(tmp_path))
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        t = Trainer(cfg)
+    assert not t.use_cl, "224^2 bf16 graph path must run NCHW"
+    assert any("channels_last" in str(x.message) for x in w)
+    t.close()
