@@ -51,6 +51,8 @@ def main():
     p.add_argument("--group-size", type=int, default=3)
     p.add_argument("--worker-fail", type=int, default=1)
     p.add_argument("--err-mode", type=str, default="rev_grad")
+    p.add_argument("--lr", type=float, default=0.01)
+    p.add_argument("--vote-granularity", type=str, default="row")
     p.add_argument("--dtype", type=str, default="bf16")
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--channels-last", type=lambda v: v.lower() in ("1","true"), default=True)
@@ -78,6 +80,8 @@ def main():
         group_size=args.group_size,
         worker_fail=args.worker_fail,
         err_mode=args.err_mode,
+        lr=args.lr,
+        vote_granularity=args.vote_granularity,
         dtype=args.dtype,
         device=args.device,
         max_steps=args.steps + args.warmup + 10,
