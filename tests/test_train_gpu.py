@@ -168,3 +168,65 @@ def test_nhwc_graph_guard_for_224_shapes(tmp_path):
     assert not t.use_cl, "224^2 bf16 graph path must run NCHW"
     assert any("channels_last" in str(x.message) for x in w)
     t.close()
+
+
+def test_gpu_geomedian_matches_cpu_fixpoint():
+    """GPU GeoMedian (fixed 24 Weiszfeld iterations, no host syncs) must land on
+    the CPU early-exit lane's fixpoint."""
+    import torch.nn as nn
+
+    from draco_amd.parallel.aggregators import GeoMedianAggregator
+    from draco_amd.parallel.comm import Communicator
+    from draco_amd.parallel.flat import FlatSpace
+
+    torch.manual_seed(0)
+    P = 7
+    payload_cpu = None
+    outs = {}
+    for dev in ("cpu", "cuda:0"):
+        device = torch.device(dev)
+        torch.manual_seed(3)
+        model = nn.Sequential(nn.Linear(200, 50), nn.Linear(50, 7)).to(device)
+        space = FlatSpace(model, 1, device)
+        comm = Communicator(0, 1, device)
+        agg = GeoMedianAggregator(comm, space, num_workers=P)
+        if payload_cpu is None:
+            payload_cpu = torch.randn(P, space.d_pad)
+            payload_cpu[2] *= 40.0  # outlier the median must resist
+        payload = space.alloc_payload(P)
+        payload.copy_(payload_cpu.to(device))
+        outs[dev] = agg.aggregate(payload, step=0).cpu()
+    diff = (outs["cpu"] - outs["cuda:0"]).abs().max()
+    scale = outs["cpu"].abs().max()
+    assert float(diff) < 1e-3 * float(scale), float(diff)
+
+
+def test_gpu_krum_matches_cpu():
+    """Device-side Krum selection (fp64 scores, gather assembly) == CPU result."""
+    import torch.nn as nn
+
+    from draco_amd.parallel.aggregators import KrumAggregator
+    from draco_amd.parallel.comm import Communicator
+    from draco_amd.parallel.flat import FlatSpace
+
+    P, s = 8, 1
+    payload_cpu = None
+    outs = {}
+    for dev in ("cpu", "cuda:0"):
+        device = torch.device(dev)
+        torch.manual_seed(5)
+        model = nn.Sequential(nn.Linear(300, 40), nn.Linear(40, 9)).to(device)
+        space = FlatSpace(model, 1, device)
+        comm = Communicator(0, 1, device)
+        agg = KrumAggregator(comm, space, num_workers=P, s=s)
+        if payload_cpu is None:
+            torch.manual_seed(9)
+            payload_cpu = torch.randn(P, space.d_pad)
+            payload_cpu[5] += 100.0  # adversary Krum must not select
+        payload = space.alloc_payload(P)
+        payload.copy_(payload_cpu.to(device))
+        outs[dev] = agg.aggregate(payload, step=0).cpu()
+    assert torch.allclose(outs["cpu"], outs["cuda:0"], atol=1e-4), \
+        float((outs["cpu"] - outs["cuda:0"]).abs().max())
+    # the adversarial row must not appear in the output
+    assert float((outs["cuda:0"] - payload_cpu[5]).abs().min()) > 1.0
