@@ -46,7 +46,22 @@ def main():
     fb.segment_weighted_mean(payload, w, seg, out_c)
     out_g = torch.zeros(space.d_pad, device=device)
     ops.segment_weighted_mean(xg, w.to(device), seg, out_g)
-    print("wmean maxdiff:", float((out_c - out_g.cpu()).abs().max()))
+    d = (out_c - out_g.cpu()).abs()
+    print("wmean maxdiff:", float(d.max()), "at", int(d.argmax()))
+    i = int(d.argmax())
+    segl = int((seg[:-1] <= i).sum()) - 1
+    print("  index", i, "segment", segl, "cpu", float(out_c[i]), "gpu", float(out_g[i]))
+    # manual torch-on-GPU reference: distinguishes kernel bug vs fallback bug
+    man = torch.zeros(space.d_pad, device=device)
+    wgpu = w.to(device)
+    for l in range(len(seg) - 1):
+        lo, hi = int(seg[l]), int(seg[l + 1])
+        man[lo:hi] = wgpu[:, l] @ xg[:, lo:hi]
+    print("  manual-gpu vs cpu:", float((man.cpu() - out_c).abs().max()),
+          " manual-gpu vs kernel:", float((man - out_g).abs().max().cpu()))
+    print("  mismatching elements:", int((d > 1e-3).sum()), "of", space.d_pad)
+    bad = (d > 1e-3).nonzero().flatten()
+    print("  first bad idxs:", bad[:10].tolist())
 
     # full iteration trace
     z_c, z_g = z.clone(), z.to(device).clone()
