@@ -621,6 +621,9 @@ __global__ void k_seg_wmean(const float *__restrict__ x, const float *__restrict
   long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
   long gstride = gridDim.x * (long)blockDim.x;
   for (; i < d; i += gstride) {
+    if (i < s_seg[0] || i >= s_seg[L]) continue;  // outside all segments: out is
+    // left untouched, matching the CPU fallback (callers zero-init the buffer;
+    // the [d, d_pad) padding tail must stay zero)
     int lo = 0, hi = L - 1;
     while (lo < hi) {
       int mid = (lo + hi + 1) >> 1;

@@ -192,6 +192,7 @@ def test_gpu_geomedian_matches_cpu_fixpoint():
         agg = GeoMedianAggregator(comm, space, num_workers=P)
         if payload_cpu is None:
             payload_cpu = torch.randn(P, space.d_pad)
+            payload_cpu[:, space.d:] = 0.0  # pad tail is always zero in production
             payload_cpu[2] *= 40.0  # outlier the median must resist
         payload = space.alloc_payload(P)
         payload.copy_(payload_cpu.to(device))
@@ -222,6 +223,7 @@ def test_gpu_krum_matches_cpu():
         if payload_cpu is None:
             torch.manual_seed(9)
             payload_cpu = torch.randn(P, space.d_pad)
+            payload_cpu[:, space.d:] = 0.0  # pad tail is always zero in production
             payload_cpu[5] += 100.0  # adversary Krum must not select
         payload = space.alloc_payload(P)
         payload.copy_(payload_cpu.to(device))
