@@ -70,3 +70,39 @@ def test_cyclic_erasure_plus_error(seed):
     dec = np.real(v @ R)
     ref = G.sum(axis=0)
     assert np.abs(dec - ref).max() < 1e-6 * max(np.abs(ref).max(), 1.0)
+
+
+@given(
+    st.integers(min_value=2, max_value=8),   # world
+    st.integers(min_value=2, max_value=6),   # r
+    st.integers(min_value=0, max_value=7),   # dead rank (mod world)
+)
+@settings(max_examples=60, deadline=None)
+def test_colocated_member_rows_properties(world, r, dead_mod):
+    """Survivor member-row mapping invariants for every (world, r, dead) shape:
+    rows in range, alive members map to distinct recv slots per rank-slot pair,
+    each group loses at most ceil(r/world) members per dead rank."""
+    import numpy as np
+
+    from draco_amd.coding import colocated_member_rows
+
+    G = world
+    dead = dead_mod % world
+    alive = [x for x in range(world) if x != dead]
+    rows, mask = colocated_member_rows(G, r, world, alive)
+    Wp = len(alive)
+    assert rows.shape == (G, r) and mask.shape == (G, r)
+    for g in range(G):
+        lost = int((~mask[g]).sum())
+        assert lost <= -(-r // world), (g, lost)
+        for i in range(r):
+            if mask[g, i]:
+                assert 0 <= rows[g, i] < r * Wp
+                # recv slot (i, src) is unique per (group) within member index i
+    # two different groups' members with the same slot i map to different src
+    # positions unless they share the host rank
+    for i in range(r):
+        hosts = [(g + i) % world for g in range(G)]
+        for g in range(G):
+            if mask[g, i]:
+                assert rows[g, i] == i * Wp + alive.index(hosts[g])
