@@ -106,3 +106,35 @@ def test_colocated_member_rows_properties(world, r, dead_mod):
         for g in range(G):
             if mask[g, i]:
                 assert rows[g, i] == i * Wp + alive.index(hosts[g])
+
+
+@given(
+    st.lists(st.integers(min_value=1, max_value=300), min_size=2, max_size=8),
+    st.integers(min_value=1, max_value=4),
+    st.floats(min_value=0.0005, max_value=2.0),
+)
+@settings(max_examples=40, deadline=None)
+def test_bucket_tiling_random_models(widths, world, bucket_mb):
+    """Bucket ranges tile [0, d_pad) top-down with aligned boundaries for any
+    layer-size profile / world / bucket size."""
+    import torch
+    import torch.nn as nn
+
+    from draco_amd.parallel.flat import ALIGN, FlatSpace
+
+    layers = []
+    prev = 7
+    for w in widths:
+        layers.append(nn.Linear(prev, w))
+        prev = w
+    space = FlatSpace(nn.Sequential(*layers), world, torch.device("cpu"))
+    buckets = space.build_buckets(bucket_mb)
+    expect_hi = space.d_pad
+    seen = set()
+    for lo, hi, idxs in buckets:
+        assert hi == expect_hi and lo < hi
+        assert lo % ALIGN == 0 or lo == 0
+        seen.update(idxs)
+        expect_hi = lo
+    assert expect_hi == 0
+    assert seen == set(range(len(space.params)))
