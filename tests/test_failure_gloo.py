@@ -26,9 +26,10 @@ def _failure_worker(rank, world, approach, kw, die_rank, die_step):
     from draco_amd.config import Config
     from draco_amd.parallel.trainer import Trainer
 
-    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
-                 approach=approach, err_mode="none", max_steps=100, eval_freq=0,
-                 log_dir="", train_dir="/tmp/draco_fail", health_timeout=1.5, **kw)
+    base = dict(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                approach=approach, err_mode="none", max_steps=100, eval_freq=0,
+                log_dir="", train_dir="/tmp/draco_fail", health_timeout=1.5)
+    cfg = Config(**{**base, **kw})
     t = Trainer(cfg)
     t.logger.stdout_every = 0
     losses = []
@@ -71,6 +72,13 @@ def test_cyclic_survives_rank_death():
 
 def test_baseline_mean_survives_rank_death():
     _run_failure(3, "baseline", dict(mode="normal", worker_fail=0))
+
+
+def test_vote_survives_rank_death_with_many_buckets():
+    """Rank dies while MANY per-bucket collectives are in flight (bucket_mb tiny):
+    the retry must abandon the orphaned bucket works cleanly and keep training."""
+    _run_failure(3, "maj_vote",
+                 dict(mode="maj_vote", group_size=3, worker_fail=1, bucket_mb=0.01))
 
 
 def _combined_worker(rank, world, die_rank, die_step):
