@@ -35,3 +35,21 @@ def test_torchrun_train_cli(tmp_path):
     # checkpoints in the reference layout (model_step_N) at both eval boundaries
     assert os.path.exists(tmp_path / "model_step_3")
     assert os.path.exists(tmp_path / "model_step_6")
+
+
+def test_torchrun_ps_topology_cli(tmp_path):
+    """Reference-parity PS topology through the CLI: 1 master + 2 workers on gloo."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "3", "--master-addr", "127.0.0.1",
+         "--master-port", str(_free_port()),
+         "-m", "draco_amd.train", "--",
+         "--topology", "ps", "--approach", "maj_vote", "--group-size", "2",
+         "--worker-fail", "0", "--err-mode", "none",
+         "--network", "FC", "--dataset", "MNIST", "--batch-size", "8",
+         "--device", "cpu", "--max-steps", "4", "--eval-freq", "2",
+         "--train-dir", str(tmp_path), "--log-dir", ""],
+        cwd=REPO, capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-3000:]
+    assert os.path.exists(tmp_path / "model_step_2")
+    assert os.path.exists(tmp_path / "model_step_4")
