@@ -28,7 +28,8 @@ def evaluate_checkpoint(path: str, device: torch.device, batches: int = 10, batc
     model = build_model(payload["network"], payload["dataset"]).to(device)
     model.load_state_dict(payload["model"])
     model.eval()
-    data = SyntheticClassification(payload["dataset"], device, seed=1234)
+    data = SyntheticClassification(payload["dataset"], device, seed=1234,
+                                   task=payload.get("synthetic_task", "means"))
     p1 = p5 = loss = 0.0
     with torch.no_grad():
         for b in range(batches):

@@ -74,6 +74,9 @@ def save_checkpoint(path: str, model, space, opt, step: int, cfg) -> None:
         "optimizer": _pack_opt_state(space, opt) if opt is not None else None,
         "network": cfg.network,
         "dataset": cfg.dataset,
+        # the held-out stream is task-dependent; the evaluator must score against
+        # the same synthetic task the checkpoint was trained on
+        "synthetic_task": getattr(cfg, "synthetic_task", "means"),
     }
     tmp = path + ".tmp"
     torch.save(payload, tmp)

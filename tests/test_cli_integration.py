@@ -53,3 +53,27 @@ def test_torchrun_ps_topology_cli(tmp_path):
     assert out.returncode == 0, out.stderr[-3000:]
     assert os.path.exists(tmp_path / "model_step_2")
     assert os.path.exists(tmp_path / "model_step_4")
+
+
+def test_evaluator_cli_once(tmp_path):
+    """`python -m draco_amd.evaluate --once` scores checkpoints in the reference
+    print format, honoring the checkpoint's synthetic task."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=16, device="cpu", lr=0.05,
+                 approach="baseline", mode="normal", worker_fail=0,
+                 synthetic_task="teacher", max_steps=10, eval_freq=0, log_dir="",
+                 train_dir=str(tmp_path))
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    for _ in range(4):
+        t.train_step()
+    t.save()
+    t.close()
+    out = subprocess.run(
+        [sys.executable, "-m", "draco_amd.evaluate", "--model-dir", str(tmp_path),
+         "--once", "--eval-batch-size", "32"],
+        cwd=REPO, capture_output=True, text=True, timeout=180)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "Prec@1:" in out.stdout and "Prec@5:" in out.stdout
