@@ -116,3 +116,35 @@ def test_vote_survives_rank_death_with_active_adversary():
         losses = res[r][0]
         assert losses[-1] < losses[0], "did not keep converging"
         assert res[r][2] == 0, "vote degenerated under combined erasure+adversary"
+
+
+def _adam_failure_worker(rank, world, die_rank, die_step):
+    """Failure recovery must re-shard ALL optimizer state (Adam moments), not just
+    params: survivors' flat buffers change size with the survivor world."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.005,
+                 approach="maj_vote", mode="maj_vote", group_size=3, worker_fail=1,
+                 err_mode="none", optimizer="adam", max_steps=100, eval_freq=0,
+                 log_dir="", train_dir="/tmp/draco_fail_adam", health_timeout=1.5)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    losses = []
+    for i in range(12):
+        if rank == die_rank and i == die_step:
+            time.sleep(0.2)
+            os._exit(0)
+        losses.append(t.train_step()["loss"])
+    h = float(t.space.flat_param.double().sum())
+    m = float(t.opt.exp_avg.double().sum())
+    v = float(t.opt.exp_avg_sq.double().sum())
+    t.close()
+    return (losses, h, m, v)
+
+
+def test_adam_state_survives_rank_death():
+    res = run_dist(_adam_failure_worker, 3, 2, 3, timeout=240.0, expect_missing={2})
+    for k in (1, 2, 3):  # params + both Adam moments identical across survivors
+        assert res[0][k] == res[1][k], f"survivor state field {k} diverged"
+    assert res[0][0][-1] < res[0][0][0]
