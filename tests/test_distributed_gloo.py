@@ -380,3 +380,31 @@ def test_checkpoint_collective_no_deadlock(tmp_path):
     import os
 
     assert os.path.exists(os.path.join(str(tmp_path), "model_step_4"))
+
+
+def _bucket_bf16_worker(rank, world):
+    """Bucketed overlap + bf16 wire: the per-bucket slice cast into the send
+    staging buffer must reproduce the whole-row bf16 exchange bit-for-bit."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    out = {}
+    for bucket_mb in (0.0, 0.01):
+        cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                     approach="maj_vote", mode="maj_vote", group_size=3, worker_fail=1,
+                     err_mode="rev_grad", compress_grad="bf16", max_steps=50,
+                     eval_freq=0, log_dir="", train_dir="/tmp/draco_bkt16",
+                     bucket_mb=bucket_mb)
+        t = Trainer(cfg)
+        t.logger.stdout_every = 0
+        for _ in range(4):
+            t.train_step()
+        out[bucket_mb] = t.space.flat_param.clone()
+        t.close()
+    assert torch.equal(out[0.0], out[0.01]), \
+        float((out[0.0] - out[0.01]).abs().max())
+    return True
+
+
+def test_bucketed_bf16_wire_bitwise_equal():
+    run_dist(_bucket_bf16_worker, 3)
