@@ -282,7 +282,15 @@ class Trainer:
         space, rebuild the decode with the dead rank's workers as erasures, and
         adopt a canonical survivor's (params, optimizer, step) — survivors may have
         stopped on different sides of the failed step's update, so state is
-        broadcast from the lowest surviving rank rather than assumed equal."""
+        broadcast from the lowest surviving rank rather than assumed equal.
+
+        Edge semantics: if the failure surfaced in a collective AFTER the step's
+        optimizer update (e.g. the checkpoint-boundary sync_buffers), the retry
+        re-runs the whole step on the adopted post-update state — that step's
+        update lands twice and one step number is skipped.  Consistent across
+        survivors (all adopt the same canonical state) and benign for training;
+        exactly-once step semantics under mid-collective failure would need an
+        update journal, which a Byzantine-tolerant trainer does not require."""
         from ..utils.checkpoint import _flat_to_params, _params_to_flat
 
         group, alive = self.health.declare_dead(dead)
