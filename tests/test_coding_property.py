@@ -138,3 +138,58 @@ def test_bucket_tiling_random_models(widths, world, bucket_mb):
         expect_hi = lo
     assert expect_hi == 0
     assert seen == set(range(len(space.params)))
+
+
+@given(
+    st.integers(min_value=2, max_value=5),   # world
+    st.integers(min_value=3, max_value=6),   # r
+    st.integers(min_value=0, max_value=31),  # adversary pattern seed
+)
+@settings(max_examples=40, deadline=None)
+def test_vote_with_forfeits_recovers_honest_mean(world, r, adv_seed):
+    """Unsharded VoteAggregator with a dead rank's members forfeited: as long as
+    each group keeps an honest majority among its ALIVE members, the aggregate is
+    exactly the mean of honest gradients."""
+    import numpy as np
+    import torch
+    import torch.nn as nn
+
+    from draco_amd.coding import colocated_member_rows
+    from draco_amd.parallel.aggregators import VoteAggregator
+    from draco_amd.parallel.comm import Communicator
+    from draco_amd.parallel.flat import FlatSpace
+
+    rng = np.random.default_rng(adv_seed)
+    dead = int(rng.integers(0, world))
+    alive = [x for x in range(world) if x != dead]
+    G = world
+    rows, mask = colocated_member_rows(G, r, world, alive)
+    torch.manual_seed(1)
+    model = nn.Linear(30, 4)
+    space = FlatSpace(model, 1, torch.device("cpu"))
+    comm = Communicator(0, 1, torch.device("cpu"))
+    # recv layout: row i*Wp + pos(src) over ALL alive (simulate post-exchange)
+    Wp = len(alive)
+    recv = torch.zeros(r * Wp, space.d_pad)
+    honest = {}
+    expected = torch.zeros(space.d_pad)
+    for g in range(G):
+        torch.manual_seed(100 + g)
+        hg = torch.randn(space.d_pad)
+        honest[g] = hg
+        expected += hg
+        alive_members = [i for i in range(r) if mask[g, i]]
+        n_alive = len(alive_members)
+        max_adv = (n_alive - 1) // 2
+        advs = set(rng.choice(alive_members, size=rng.integers(0, max_adv + 1),
+                              replace=False).tolist()) if max_adv > 0 else set()
+        for i in alive_members:
+            recv[rows[g, i]] = hg * (-100.0 if i in advs else 1.0)
+    expected /= G
+    agg = VoteAggregator(comm, space, group_size=r, atol=0.0,
+                         member_rows=rows, member_mask=mask)
+    # feed the pre-exchanged rows directly (world-1 comm: exchanged() is a view,
+    # so hand aggregate() a payload shaped (r*Wp, d_pad))
+    out = agg.aggregate(recv, step=0)
+    assert torch.allclose(out, expected, atol=1e-5), \
+        float((out - expected).abs().max())
