@@ -193,3 +193,62 @@ def test_vote_with_forfeits_recovers_honest_mean(world, r, adv_seed):
     out = agg.aggregate(recv, step=0)
     assert torch.allclose(out, expected, atol=1e-5), \
         float((out - expected).abs().max())
+
+
+@given(
+    st.integers(min_value=3, max_value=4),   # world0
+    st.integers(min_value=1, max_value=2),   # L (workers per rank)
+    st.integers(min_value=0, max_value=97),  # seed
+)
+@settings(max_examples=25, deadline=None)
+def test_cyclic_survivor_mapping_decodes_exact(world0, L, seed):
+    """CyclicAggregator with a dead original rank: recv rows in survivor-world
+    coordinates + erased workers as known-bad must still decode the exact sum,
+    for every (world0, L, dead) geometry within capability (L erasures <= s)."""
+    import numpy as np
+    import torch
+    import torch.nn as nn
+
+    from draco_amd import ops
+    from draco_amd.coding import build_cyclic_code
+    from draco_amd.parallel.aggregators import CyclicAggregator
+    from draco_amd.parallel.comm import Communicator
+    from draco_amd.parallel.flat import FlatSpace
+
+    rng = np.random.default_rng(seed)
+    n = L * world0
+    s = L  # one dead rank erases exactly L workers; stay within the locator budget
+    if n < 2 * s + 2:
+        return  # geometry outside code constraints
+    dead = int(rng.integers(0, world0))
+    alive = [x for x in range(world0) if x != dead]
+    code = build_cyclic_code(n, s)
+    torch.manual_seed(2)
+    model = nn.Linear(40, 5)
+    space = FlatSpace(model, 1, torch.device("cpu"))
+    comm = Communicator(0, 1, torch.device("cpu"))
+    agg = CyclicAggregator(comm, space, code, workers_per_rank=L,
+                           world0=world0, alive=alive)
+
+    def sub_grad(j):
+        torch.manual_seed(700 + j)
+        return torch.randn(space.d_pad)
+
+    Wp = len(alive)
+    recv = torch.zeros(2 * L * Wp, space.d_pad)
+    for w in range(n):
+        l, src = w // world0, w % world0
+        if src == dead:
+            continue  # erased: its rows never arrive
+        sup = code.support[w]
+        grads = torch.stack([sub_grad(int(j)) for j in sup])
+        wre = torch.tensor(np.real(code.W[w, sup]), dtype=torch.float32)
+        wim = torch.tensor(np.imag(code.W[w, sup]), dtype=torch.float32)
+        enc = torch.zeros(2, space.d_pad)
+        ops.cyclic_encode(grads, wre, wim, enc)
+        pos = alive.index(src)
+        recv[(2 * l) * Wp + pos] = enc[0]
+        recv[(2 * l + 1) * Wp + pos] = enc[1]
+    out = agg.aggregate(recv, step=int(rng.integers(0, 1000)))
+    ref = torch.stack([sub_grad(j) for j in range(n)]).sum(0) / n
+    assert torch.allclose(out, ref, atol=1e-4), float((out - ref).abs().max())
