@@ -408,3 +408,46 @@ def _bucket_bf16_worker(rank, world):
 
 def test_bucketed_bf16_wire_bitwise_equal():
     run_dist(_bucket_bf16_worker, 3)
+
+
+def _err_mode_worker(rank, world, approach, err_mode, kw):
+    """Every reference err mode must be excluded by the coded decodes end-to-end."""
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    out = {}
+    for mode in (err_mode, "none"):
+        base = dict(network="FC", dataset="MNIST", batch_size=4, device="cpu",
+                    lr=0.05, approach=approach, err_mode=mode, max_steps=50,
+                    eval_freq=0, log_dir="", train_dir="/tmp/draco_em")
+        t = Trainer(Config(**{**base, **kw}))
+        t.logger.stdout_every = 0
+        for _ in range(5):
+            t.train_step()
+        out[mode] = t.space.flat_param.clone()
+        t.close()
+    # the decode removes the adversary entirely.  maj_vote: the winner is an
+    # honest member's row, so attacked == clean BIT-FOR-BIT (CPU bitwise vote).
+    # cyclic: the attacked run decodes through a subset recombination vector
+    # (clean takes the zero-syndrome fast path) — identical in real arithmetic,
+    # different fp rounding, so compare to fp accumulation tolerance.
+    if approach == "maj_vote":
+        assert torch.equal(out[err_mode], out["none"]), \
+            float((out[err_mode] - out["none"]).abs().max())
+    else:
+        diff = float((out[err_mode] - out["none"]).abs().max())
+        scale = float(out["none"].abs().max())
+        assert diff <= 1e-3 * max(scale, 1e-6), diff
+    return True
+
+
+@pytest.mark.parametrize("err_mode", ["constant", "rev_grad"])
+def test_vote_excludes_all_err_modes(err_mode):
+    run_dist(_err_mode_worker, 3, "maj_vote", err_mode,
+             dict(mode="maj_vote", group_size=3, worker_fail=1))
+
+
+@pytest.mark.parametrize("err_mode", ["constant", "rev_grad", "gauss"])
+def test_cyclic_excludes_all_err_modes(err_mode):
+    run_dist(_err_mode_worker, 2, "cyclic", err_mode,
+             dict(mode="cyclic", worker_fail=1, workers_per_rank=2))
