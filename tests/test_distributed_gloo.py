@@ -451,3 +451,48 @@ def test_vote_excludes_all_err_modes(err_mode):
 def test_cyclic_excludes_all_err_modes(err_mode):
     run_dist(_err_mode_worker, 2, "cyclic", err_mode,
              dict(mode="cyclic", worker_fail=1, workers_per_rank=2))
+
+
+def _save_worker(rank, world, ckdir):
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                 approach="maj_vote", mode="maj_vote", group_size=3, worker_fail=1,
+                 max_steps=50, eval_freq=4, log_dir="", train_dir=ckdir)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    for _ in range(4):
+        t.train_step()
+    h = float(t.space.flat_param.double().sum())
+    t.close()
+    return h
+
+
+def _resume_worker(rank, world, ckdir):
+    from draco_amd.config import Config
+    from draco_amd.parallel.trainer import Trainer
+
+    cfg = Config(network="FC", dataset="MNIST", batch_size=4, device="cpu", lr=0.05,
+                 approach="maj_vote", mode="maj_vote", group_size=3, worker_fail=1,
+                 max_steps=50, eval_freq=0, log_dir="", train_dir=ckdir,
+                 checkpoint_step=4)
+    t = Trainer(cfg)
+    t.logger.stdout_every = 0
+    h0 = float(t.space.flat_param.double().sum())
+    assert t.step_num == 4
+    r = t.train_step()  # must keep training from the restored state
+    t.close()
+    return (h0, r["loss"])
+
+
+def test_checkpoint_resume_across_world_sizes(tmp_path):
+    """model_step_N checkpoints are world-size independent: save from a world-2 run
+    (different shard padding) and resume at world 3 with identical parameters."""
+    ckdir = str(tmp_path)
+    saved = run_dist(_save_worker, 2, ckdir)
+    assert saved[0] == saved[1]
+    resumed = run_dist(_resume_worker, 3, ckdir)
+    for r in range(3):
+        assert abs(resumed[r][0] - saved[0]) < 1e-9, (resumed[r][0], saved[0])
+        assert resumed[r][1] == resumed[0][1]
